@@ -1,0 +1,188 @@
+"""Gaussian random field initialization with prescribed power spectra.
+
+Analogue of reference pystella/fourier/rayleigh.py:35-395.  Mode
+amplitudes are Rayleigh-distributed (amp = √(−ln u₀)) with uniform
+phases; the WKB variant initializes (f, ḟ) pairs for Klein-Gordon
+fields in conformal FLRW.  Random numbers come from torch's
+counter-based Philox generator (per-rank seeds), replacing
+pyopencl.clrandom Threefry (reference rayleigh.py:154).
+
+For real fields the kz=0 and kz=Nyquist planes are made exactly
+Hermitian-symmetric (vectorized analogue of reference
+``make_hermitian``, rayleigh.py:35-55) and corner-mode imaginary parts
+are zeroed, so c2r transforms see consistent data.
+"""
+
+from __future__ import annotations
+
+import numpy as np
+import torch
+
+__all__ = ["RayleighGenerator", "make_hermitian"]
+
+
+def make_hermitian(fk):
+    """Enforce Hermitian symmetry on the kz ∈ {0, Nyquist} planes of an
+    r2c half-space array (torch tensor, in place)."""
+    Nx, Ny, NKz = fk.shape
+    Nz = 2 * (NKz - 1)
+    dev = fk.device
+    ii = torch.arange(Nx, device=dev).view(-1, 1)
+    jj = torch.arange(Ny, device=dev).view(1, -1)
+    ni = (-ii) % Nx
+    nj = (-jj) % Ny
+    keep = (ii < ni) | ((ii == ni) & (jj <= nj))
+    for k in {0, Nz // 2}:
+        if k >= NKz:
+            continue
+        P = fk[:, :, k]
+        partner = P[ni.expand(Nx, Ny), nj.expand(Nx, Ny)]
+        fk[:, :, k] = torch.where(keep, P, partner.conj())
+    return fk
+
+
+class RayleighGenerator:
+    """Draws random field realizations with prescribed power spectra
+    (reference rayleigh.py:57-395).
+
+    :arg fft: an FFT object from :func:`~pystella_amd.DFT`.
+    :arg dk: 3-tuple momentum-space spacing.
+    :arg volume: physical box volume.
+    :arg seed: RNG seed (make it rank-dependent for distributed runs,
+        as the reference example does: scalar_preheating.py:227).
+    """
+
+    def __init__(self, context_or_fft=None, fft=None, dk=None, volume=None,
+                 seed=13298):
+        if fft is None:
+            fft = context_or_fft
+        self.fft = fft
+        self.volume = volume
+        dev = fft.fk.device
+        self.device = dev
+
+        sub_k = [fft.sub_k[n].cpu().numpy()
+                 for n in ("momenta_x", "momenta_y", "momenta_z")]
+        kvecs = np.meshgrid(*sub_k, indexing="ij", sparse=False)
+        self.kmags = np.sqrt(sum((dki * ki)**2
+                                 for dki, ki in zip(dk, kvecs)))
+        self._kmags_t = torch.as_tensor(self.kmags, device=dev)
+        self.gen = torch.Generator(device=dev)
+        self.gen.manual_seed(int(seed))
+        self.cdtype = fft.fk.dtype
+
+    # ------------------------------------------------------------------
+    def _uniform(self, n_sets):
+        shape = (n_sets,) + self.kmags.shape
+        u = torch.rand(shape, dtype=torch.float64, device=self.device,
+                       generator=self.gen)
+        return u.clamp_(min=1e-300)
+
+    def _post_process(self, fk):
+        if self.fft.is_real:
+            make_hermitian(fk)
+            self.fft.zero_corner_modes(fk, only_imag=True)
+        return fk
+
+    def _ps_wrapper(self, ps_func, wk, kmags):
+        """Evaluate a power spectrum, zeroing the homogeneous mode if
+        this rank holds it (reference rayleigh.py:174-185)."""
+        if isinstance(wk, torch.Tensor):
+            wk_np = wk.cpu().numpy()
+        else:
+            wk_np = np.asarray(wk)
+        found_zero = kmags.flat[0] == 0. and np.all(
+            np.unravel_index(0, kmags.shape) == (0, 0, 0))
+        wk_np = wk_np.copy()
+        if found_zero and kmags[0, 0, 0] == 0.:
+            wk_np[0, 0, 0] = wk_np[0, 0, 1]
+        power = ps_func(wk_np)
+        power = np.asarray(power, dtype=np.float64) * np.ones_like(wk_np)
+        if found_zero and kmags[0, 0, 0] == 0.:
+            power[0, 0, 0] = 0.
+        return torch.as_tensor(power, device=self.device)
+
+    # ------------------------------------------------------------------
+    def generate(self, queue=None, random=True,
+                 field_ps=lambda kmag: 1 / 2 / kmag, norm=1,
+                 window=lambda kmag: 1.):
+        """Generate Fourier modes with power spectrum ``field_ps``
+        (reference rayleigh.py:185-227)."""
+        amplitude_sq = norm / self.volume
+        rands = self._uniform(2)
+        if not random:
+            rands[0] = np.exp(-1.)
+
+        f_power = (amplitude_sq
+                   * torch.as_tensor(window(self.kmags) ** 2
+                                     * np.ones_like(self.kmags),
+                                     device=self.device)
+                   * self._ps_wrapper(field_ps, self.kmags, self.kmags))
+
+        amp = torch.sqrt(-torch.log(rands[0]))
+        phs = torch.exp(2j * np.pi * rands[1])
+        fk = phs * amp * torch.sqrt(f_power)
+        return self._post_process(fk.to(self.cdtype))
+
+    def generate_WKB(self, queue=None, random=True,
+                     field_ps=lambda wk: 1 / 2 / wk, norm=1,
+                     omega_k=lambda kmag: kmag, hubble=0.,
+                     window=lambda kmag: 1.):
+        """Generate (f_k, ḟ_k) pairs via the WKB approximation
+        (reference rayleigh.py:325-373)."""
+        amplitude_sq = norm / self.volume
+        rands = self._uniform(4)
+        if not random:
+            rands[0] = rands[2] = np.exp(-1.)
+
+        wk = np.asarray(omega_k(self.kmags)) * np.ones_like(self.kmags)
+        f_power = (amplitude_sq
+                   * torch.as_tensor(window(self.kmags) ** 2
+                                     * np.ones_like(self.kmags),
+                                     device=self.device)
+                   * self._ps_wrapper(field_ps, wk, self.kmags))
+
+        amp1 = torch.sqrt(-torch.log(rands[0]))
+        amp2 = torch.sqrt(-torch.log(rands[2]))
+        phs1 = torch.exp(2j * np.pi * rands[1])
+        phs2 = torch.exp(2j * np.pi * rands[3])
+        sqrtp = torch.sqrt(f_power)
+        L = phs1 * amp1 * sqrtp
+        R = phs2 * amp2 * sqrtp
+        s2 = np.sqrt(2.)
+        fk = (L + R) / s2
+        wk_t = torch.as_tensor(wk, device=self.device)
+        dfk = 1j * wk_t * (L - R) / s2 - hubble * fk
+
+        fk = self._post_process(fk.to(self.cdtype))
+        dfk = self._post_process(dfk.to(self.cdtype))
+        return fk, dfk
+
+    # ------------------------------------------------------------------
+    def init_field(self, fx, queue=None, **kwargs):
+        fk = self.generate(**kwargs)
+        self.fft.idft(fk, fx)
+
+    def init_WKB_fields(self, fx, dfx, queue=None, **kwargs):
+        fk, dfk = self.generate_WKB(**kwargs)
+        self.fft.idft(fk, fx)
+        self.fft.idft(dfk, dfx)
+
+    def init_transverse_vector(self, projector, vector, queue=None, **kwargs):
+        vector_k = torch.empty((3,) + tuple(self.fft.shape(True)),
+                               dtype=self.cdtype, device=self.device)
+        for mu in range(3):
+            vector_k[mu] = self.generate(**kwargs)
+        projector.transversify(vector=vector_k)
+        for mu in range(3):
+            self.fft.idft(vector_k[mu], vector[mu])
+
+    def init_vector_from_pol(self, projector, vector, plus_ps, minus_ps,
+                             queue=None, **kwargs):
+        plus_k = self.generate(field_ps=plus_ps, **kwargs)
+        minus_k = self.generate(field_ps=minus_ps, **kwargs)
+        vector_k = torch.empty((3,) + tuple(self.fft.shape(True)),
+                               dtype=self.cdtype, device=self.device)
+        projector.pol_to_vec(plus=plus_k, minus=minus_k, vector=vector_k)
+        for mu in range(3):
+            self.fft.idft(vector_k[mu], vector[mu])

@@ -1,0 +1,171 @@
+"""Run output: appendable time-series datasets + provenance capture.
+
+Analogue of reference pystella/output.py:52-181 (``OutputFile`` over
+h5py).  This image ships no h5py, so the default backend is a
+self-contained directory store with the same append-mode semantics:
+
+* ``out.output("energy", t=..., a=..., **components)`` appends one row
+  per call to each named dataset under the group;
+* run provenance (hostname, device, argv, package versions, git
+  revisions, the run script's text) is captured as attributes, matching
+  reference output.py:98-155;
+* everything is written as ``.npy`` stacks + a JSON attribute file, and
+  can be read back with :class:`OutputFile`'s ``read`` helpers.
+
+If h5py is importable, an HDF5 file with the same layout is written
+instead (same group/dataset names and attrs).
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import socket
+import sys
+import time
+
+import numpy as np
+
+__all__ = ["OutputFile"]
+
+
+def get_versions(dependencies):
+    versions = {}
+    for dep in dependencies:
+        try:
+            mod = __import__(dep)
+            versions[dep] = str(getattr(mod, "__version__", "unknown"))
+        except ImportError:
+            versions[dep] = "not installed"
+    return versions
+
+
+def _git_rev(path):
+    import subprocess
+    try:
+        return subprocess.check_output(
+            ["git", "rev-parse", "HEAD"], cwd=path,
+            stderr=subprocess.DEVNULL).decode().strip()
+    except Exception:
+        return None
+
+
+class _DirStore:
+    """Append-mode dataset store over a directory of .npy stacks."""
+
+    def __init__(self, path):
+        self.path = path
+        os.makedirs(path, exist_ok=True)
+        self._cache = {}
+
+    def set_attrs(self, attrs):
+        with open(os.path.join(self.path, "attrs.json"), "w") as f:
+            json.dump(attrs, f, indent=1, default=str)
+
+    def append(self, group, name, value):
+        key = (group, name)
+        arr = np.asarray(value)
+        buf = self._cache.setdefault(key, [])
+        buf.append(arr)
+        gdir = os.path.join(self.path, group)
+        os.makedirs(gdir, exist_ok=True)
+        np.save(os.path.join(gdir, name + ".npy"),
+                np.stack(self._cache[key]))
+
+    def read(self, group, name):
+        return np.load(os.path.join(self.path, group, name + ".npy"))
+
+    def close(self):
+        pass
+
+
+class _H5Store:
+    def __init__(self, path):
+        import h5py
+        self.f = h5py.File(path, "a")
+
+    def set_attrs(self, attrs):
+        for k, v in attrs.items():
+            try:
+                self.f.attrs[k] = v
+            except TypeError:
+                self.f.attrs[k] = str(v)
+
+    def append(self, group, name, value):
+        arr = np.asarray(value)
+        g = self.f.require_group(group)
+        if name not in g:
+            g.create_dataset(name, shape=(0,) + arr.shape,
+                             maxshape=(None,) + arr.shape,
+                             dtype=arr.dtype)
+        ds = g[name]
+        ds.resize(ds.shape[0] + 1, axis=0)
+        ds[-1] = arr
+
+    def read(self, group, name):
+        return np.asarray(self.f[group][name])
+
+    def close(self):
+        self.f.close()
+
+
+class OutputFile:
+    """Appendable run output with provenance attributes
+    (reference output.py:52-181)."""
+
+    def __init__(self, ctx=None, name=None, runfile=None, suffix="",
+                 **kwargs):
+        if name is None:
+            stamp = time.strftime("%Y-%m-%d-%H%M%S")
+            name = f"output-{stamp}{suffix}"
+
+        attrs = {
+            "hostname": socket.gethostname(),
+            "argv": " ".join(sys.argv),
+            "created": time.strftime("%Y-%m-%d %H:%M:%S"),
+            "versions": get_versions(
+                ["numpy", "scipy", "torch", "pystella_amd"]),
+        }
+        try:
+            import torch
+            if torch.cuda.is_available():
+                attrs["device"] = torch.cuda.get_device_name(0)
+        except Exception:
+            pass
+        rev = _git_rev(os.path.dirname(os.path.abspath(
+            runfile or __file__)))
+        if rev:
+            attrs["git_rev"] = rev
+        if runfile is not None:
+            try:
+                with open(runfile) as f:
+                    attrs["runfile_text"] = f.read()
+            except OSError:
+                pass
+        attrs.update({k: str(v) for k, v in kwargs.items()})
+
+        try:
+            import h5py  # noqa: F401
+            self.store = _H5Store(name + ".h5")
+            self.filename = name + ".h5"
+        except ImportError:
+            self.store = _DirStore(name)
+            self.filename = name
+        self.store.set_attrs(attrs)
+
+    def output(self, group, **datasets):
+        """Append one row to each named dataset in ``group``."""
+        for key, val in datasets.items():
+            self.store.append(group, key, val)
+
+    def read(self, group, name):
+        return self.store.read(group, name)
+
+    def close(self):
+        self.store.close()
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *exc):
+        self.close()
