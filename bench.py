@@ -1,0 +1,157 @@
+"""Flagship benchmark: scalar-preheating RK4 step throughput.
+
+Driver contract (see repo instructions): measures the BASELINE.json
+metric — Msite-updates/sec of the scalar_preheating hot loop (2 scalar
+fields, fp64, halo 2, 512³ global grid, LowStorageRK54 = 4th-order RK)
+— on N GPUs of one node, strong scaling (fixed 512³ total grid).
+
+Each timed step is the full reference hot loop
+(reference examples/scalar_preheating.py:258-271): per RK stage, the
+fused stage kernel, the host Friedmann update, halo exchange +
+fused grad/lap stencil, and the ⟨ρ⟩/⟨P⟩ energy reduction + allreduce.
+Synthetic random-init fields (no dataset exists for this workload).
+"""
+
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+
+import pystella_amd as ps
+from pystella_amd.sectors import get_rho_and_p
+
+
+def proc_shape_for(n):
+    shapes = {1: (1, 1, 1), 2: (2, 1, 1), 4: (2, 2, 1), 8: (2, 2, 2),
+              3: (3, 1, 1), 6: (3, 2, 1)}
+    if n in shapes:
+        return shapes[n]
+    return (n, 1, 1)
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=20)
+    parser.add_argument("--warmup", type=int, default=5)
+    parser.add_argument("--grid", type=int, default=512)
+    parser.add_argument("--halo", type=int, default=2)
+    parser.add_argument("--device", default=None)
+    p = parser.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    if world_size > 1:
+        ps.init_distributed()
+    n_gpus = max(p.gpus, world_size)
+
+    if p.device is not None:
+        device = torch.device(p.device)
+    elif torch.cuda.is_available():
+        device = ps.choose_device()
+    else:
+        device = torch.device("cpu")
+    on_gpu = device.type == "cuda"
+
+    grid_shape = (p.grid,) * 3
+    grid_size = float(np.prod(grid_shape))
+    h = p.halo
+    box = (5., 5., 5.)
+    dx = tuple(L / N for L, N in zip(box, grid_shape))
+    dt = 0.1 * min(dx)
+    nscalars = 2
+    mphi, mpl, gsq = 1.2e-6, 1.0, 2.5e-7
+
+    proc_shape = proc_shape_for(world_size if world_size > 1 else 1)
+    decomp = ps.DomainDecomposition(proc_shape, h, grid_shape=grid_shape)
+    rank_shape = decomp.rank_shape
+    pad = tuple(n + 2 * h for n in rank_shape)
+
+    def potential(f):
+        phi, chi = f[0], f[1]
+        return (mphi**2 / 2 * phi**2 + gsq / 2 * phi**2 * chi**2) / mphi**2
+
+    sector = ps.ScalarSector(nscalars, potential=potential)
+    stepper = ps.LowStorageRK54([sector], halo_shape=h,
+                                rank_shape=rank_shape, dt=dt)
+    derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=rank_shape)
+    reduce_energy = ps.Reduction(
+        decomp, sector, halo_shape=h, callback=get_rho_and_p,
+        rank_shape=rank_shape, grid_size=grid_size)
+
+    gen = torch.Generator(device="cpu").manual_seed(7 + decomp.rank)
+    f = (0.193 + 1e-3 * torch.rand((nscalars,) + pad, dtype=torch.float64,
+                                   generator=gen)).to(device)
+    dfdt = (-0.142 + 1e-3 * torch.rand((nscalars,) + pad,
+                                       dtype=torch.float64,
+                                       generator=gen)).to(device)
+    lap_f = torch.zeros((nscalars,) + tuple(rank_shape),
+                        dtype=torch.float64, device=device)
+
+    energy = None
+
+    def compute_energy(a):
+        derivs(fx=f, lap=lap_f)
+        return reduce_energy(f=f, dfdt=dfdt, lap_f=lap_f, a=np.array(a))
+
+    energy = compute_energy(1.)
+    expand = ps.Expansion(energy["total"], ps.LowStorageRK54, mpl=mpl)
+
+    def step():
+        nonlocal energy
+        for s in range(stepper.num_stages):
+            stepper(s, a=expand.a, hubble=expand.hubble,
+                    f=f, dfdt=dfdt, lap_f=lap_f)
+            expand.step(s, energy["total"], energy["pressure"], dt)
+            energy = compute_energy(expand.a)
+
+    def sync():
+        if on_gpu:
+            torch.cuda.synchronize()
+        decomp.barrier()
+
+    for _ in range(p.warmup):
+        step()
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(p.steps):
+        step()
+    sync()
+    elapsed = time.perf_counter() - t0
+    # max over ranks
+    elapsed = float(decomp.allreduce(elapsed, op="max"))
+
+    ms_per_step = elapsed / p.steps * 1e3
+    msites = grid_size * p.steps / elapsed / 1e6
+
+    if decomp.rank == 0:
+        print(json.dumps({
+            "metric": "Msite-updates/sec, scalar-preheating 512^3 RK4",
+            "value": msites,
+            "unit": "Msites/s",
+            "n_gpus": n_gpus,
+            "steps": p.steps,
+            "warmup": p.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "fp64",
+            "data": "synthetic random-init fields",
+            "config": {
+                "model": "scalar_preheating",
+                "grid_shape": list(grid_shape),
+                "halo": h,
+                "nscalars": nscalars,
+                "stepper": "LowStorageRK54 (4th order, 5 stages)",
+                "global_batch": None,
+                "seq_len": None,
+                "parallelism": f"decomp3d{list(proc_shape)}",
+            },
+        }), flush=True)
+
+
+if __name__ == "__main__":
+    main()

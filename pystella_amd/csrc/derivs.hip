@@ -1,0 +1,261 @@
+// Hand-written CDNA4 (gfx950 / MI355X) finite-difference stencil kernels.
+//
+// MI355X-native re-design of the reference's loopy-generated stencil
+// kernels (reference: pystella/derivs.py:301-337, pystella/stencil.py:103-141
+// "StreamingStencil").  Design:
+//
+//  * x-marching: each 256-thread block owns a (4 y × 64 z) tile and walks
+//    the full x extent, keeping the 2H+1 x-planes needed by the x-stencil
+//    in a register ring — the expensive (large-stride) axis never re-reads
+//    HBM.  y/z neighbors are read from the current plane; consecutive
+//    lanes read consecutive z addresses (fully coalesced), so these hit
+//    L1/L2.
+//  * fused grad+lap: one pass reads f once and emits all four outputs,
+//    reusing each neighbor load for both the first- and second-derivative
+//    coefficient (the reference fuses the same way: derivs.py:334-337).
+//  * fp64 throughout; the kernels are HBM-bandwidth bound by design.
+//
+// Centered-difference coefficients of truncation order 2H
+// (reference derivs.py:127-131, 160-165; standard published tables).
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+namespace {
+
+template <int H> struct FD;
+template <> struct FD<1> {
+    __device__ static constexpr double g(int s) {
+        constexpr double c[1] = {1. / 2};
+        return c[s - 1];
+    }
+    __device__ static constexpr double l(int s) {
+        constexpr double c[2] = {-2., 1.};
+        return c[s];
+    }
+};
+template <> struct FD<2> {
+    __device__ static constexpr double g(int s) {
+        constexpr double c[2] = {8. / 12, -1. / 12};
+        return c[s - 1];
+    }
+    __device__ static constexpr double l(int s) {
+        constexpr double c[3] = {-30. / 12, 16. / 12, -1. / 12};
+        return c[s];
+    }
+};
+template <> struct FD<3> {
+    __device__ static constexpr double g(int s) {
+        constexpr double c[3] = {45. / 60, -9. / 60, 1. / 60};
+        return c[s - 1];
+    }
+    __device__ static constexpr double l(int s) {
+        constexpr double c[4] = {-490. / 180, 270. / 180, -27. / 180,
+                                 2. / 180};
+        return c[s];
+    }
+};
+template <> struct FD<4> {
+    __device__ static constexpr double g(int s) {
+        constexpr double c[4] = {672. / 840, -168. / 840, 32. / 840,
+                                 -3. / 840};
+        return c[s - 1];
+    }
+    __device__ static constexpr double l(int s) {
+        constexpr double c[5] = {-14350. / 5040, 8064. / 5040,
+                                 -1008. / 5040, 128. / 5040, -9. / 5040};
+        return c[s];
+    }
+};
+
+// Block geometry: 64 lanes along z (one wave) x BY rows along y.
+constexpr int BZ = 64;
+constexpr int BY = 4;
+
+// Fused gradient/Laplacian kernel, x-marching with a register ring.
+// f is halo-padded (nx+2H, ny+2H, nz+2H) per field; outputs unpadded.
+template <int H, bool LAP, bool GRAD>
+__global__ __launch_bounds__(BZ * BY) void gradlap_knl(
+    const double *__restrict__ f, double *__restrict__ lap,
+    double *__restrict__ pdx, double *__restrict__ pdy,
+    double *__restrict__ pdz, int nx, int ny, int nz,
+    double inv_dx, double inv_dy, double inv_dz,
+    double inv_dx2, double inv_dy2, double inv_dz2)
+{
+    const int k = blockIdx.x * BZ + (threadIdx.x % BZ);
+    const int j = blockIdx.y * BY + (threadIdx.x / BZ);
+    const int fld = blockIdx.z;
+    if (k >= nz || j >= ny) return;
+
+    const int64_t psz = nz + 2 * H;
+    const int64_t psy = ny + 2 * H;
+    const int64_t sx = psy * psz;                 // padded x-plane stride
+    const int64_t pvol = (nx + 2 * H) * sx;
+    const int64_t uvol = (int64_t)nx * ny * nz;
+
+    // pointer to (x=-H, y=j, z=k) of this field
+    const double *fp = f + (int64_t)fld * pvol + ((int64_t)(j + H)) * psz
+                       + (k + H);
+    double *outl = LAP ? lap + (int64_t)fld * uvol + (int64_t)j * nz + k
+                       : nullptr;
+    double *outx = GRAD ? pdx + (int64_t)fld * uvol + (int64_t)j * nz + k
+                        : nullptr;
+    double *outy = GRAD ? pdy + (int64_t)fld * uvol + (int64_t)j * nz + k
+                        : nullptr;
+    double *outz = GRAD ? pdz + (int64_t)fld * uvol + (int64_t)j * nz + k
+                        : nullptr;
+    const int64_t so = (int64_t)ny * nz;          // unpadded x stride
+
+    // register ring r[p] holds f at x = i - H + p (center plane p = H)
+    double r[2 * H + 1];
+#pragma unroll
+    for (int p = 0; p < 2 * H; ++p) r[p] = fp[(int64_t)p * sx];
+
+    for (int i = 0; i < nx; ++i) {
+        r[2 * H] = fp[(int64_t)(i + 2 * H) * sx];
+        const double c = r[H];
+        const double *cp = fp + (int64_t)(i + H) * sx;  // center plane
+
+        double lap_acc = 0., gx = 0., gy = 0., gz = 0.;
+        if (LAP)
+            lap_acc = FD<H>::l(0) * c * (inv_dx2 + inv_dy2 + inv_dz2);
+
+#pragma unroll
+        for (int s = 1; s <= H; ++s) {
+            const double xm = r[H - s], xp = r[H + s];
+            const double ym = cp[-(int64_t)s * psz], yp = cp[(int64_t)s * psz];
+            const double zm = cp[-s], zp = cp[s];
+            if (LAP)
+                lap_acc += FD<H>::l(s) * ((xp + xm) * inv_dx2
+                                          + (yp + ym) * inv_dy2
+                                          + (zp + zm) * inv_dz2);
+            if (GRAD) {
+                gx += FD<H>::g(s) * (xp - xm);
+                gy += FD<H>::g(s) * (yp - ym);
+                gz += FD<H>::g(s) * (zp - zm);
+            }
+        }
+        if (LAP) outl[(int64_t)i * so] = lap_acc;
+        if (GRAD) {
+            outx[(int64_t)i * so] = gx * inv_dx;
+            outy[(int64_t)i * so] = gy * inv_dy;
+            outz[(int64_t)i * so] = gz * inv_dz;
+        }
+        // rotate the ring
+#pragma unroll
+        for (int p = 0; p < 2 * H; ++p) r[p] = r[p + 1];
+    }
+}
+
+// Single-axis first derivative (optionally accumulating, for divergence).
+// AXIS: 0=x, 1=y, 2=z.
+template <int H, int AXIS, bool ACCUM>
+__global__ __launch_bounds__(BZ * BY) void pd_knl(
+    const double *__restrict__ f, double *__restrict__ out,
+    int nx, int ny, int nz, double inv_d)
+{
+    const int k = blockIdx.x * BZ + (threadIdx.x % BZ);
+    const int j = blockIdx.y * BY + (threadIdx.x / BZ);
+    const int fld = blockIdx.z;
+    if (k >= nz || j >= ny) return;
+
+    const int64_t psz = nz + 2 * H;
+    const int64_t psy = ny + 2 * H;
+    const int64_t sx = psy * psz;
+    const int64_t pvol = (nx + 2 * H) * sx;
+    const int64_t uvol = (int64_t)nx * ny * nz;
+    const int64_t nstride = (AXIS == 0) ? sx : (AXIS == 1) ? psz : 1;
+
+    const double *fp = f + (int64_t)fld * pvol + ((int64_t)(j + H)) * psz
+                       + (k + H);
+    double *op = out + (int64_t)fld * uvol + (int64_t)j * nz + k;
+    const int64_t so = (int64_t)ny * nz;
+
+    for (int i = 0; i < nx; ++i) {
+        const double *cp = fp + (int64_t)(i + H) * sx;
+        double g = 0.;
+#pragma unroll
+        for (int s = 1; s <= H; ++s)
+            g += FD<H>::g(s) * (cp[(int64_t)s * nstride]
+                                - cp[-(int64_t)s * nstride]);
+        g *= inv_d;
+        if (ACCUM)
+            op[(int64_t)i * so] += g;
+        else
+            op[(int64_t)i * so] = g;
+    }
+}
+
+inline dim3 tile_grid(int ny, int nz, int nf)
+{
+    return dim3((nz + BZ - 1) / BZ, (ny + BY - 1) / BY, nf);
+}
+
+}  // namespace
+
+#define DISPATCH_H(H_, ...)                                              \
+    switch (H_) {                                                        \
+    case 1: { constexpr int H = 1; __VA_ARGS__; break; }                 \
+    case 2: { constexpr int H = 2; __VA_ARGS__; break; }                 \
+    case 3: { constexpr int H = 3; __VA_ARGS__; break; }                 \
+    case 4: { constexpr int H = 4; __VA_ARGS__; break; }                 \
+    default: return 1;                                                   \
+    }
+
+extern "C" int pystella_gradlap(
+    const double *f, double *lap, double *pdx, double *pdy, double *pdz,
+    int h, int nx, int ny, int nz, int nf,
+    double dx, double dy, double dz, void *stream_)
+{
+    hipStream_t stream = (hipStream_t)stream_;
+    const dim3 grid = tile_grid(ny, nz, nf);
+    const dim3 block(BZ * BY);
+    const double ix = 1. / dx, iy = 1. / dy, iz = 1. / dz;
+    const double ix2 = ix * ix, iy2 = iy * iy, iz2 = iz * iz;
+    const bool do_lap = lap != nullptr;
+    const bool do_grad = pdx != nullptr;
+
+    DISPATCH_H(h, {
+        if (do_lap && do_grad)
+            hipLaunchKernelGGL((gradlap_knl<H, true, true>), grid, block, 0,
+                               stream, f, lap, pdx, pdy, pdz, nx, ny, nz,
+                               ix, iy, iz, ix2, iy2, iz2);
+        else if (do_lap)
+            hipLaunchKernelGGL((gradlap_knl<H, true, false>), grid, block, 0,
+                               stream, f, lap, pdx, pdy, pdz, nx, ny, nz,
+                               ix, iy, iz, ix2, iy2, iz2);
+        else
+            hipLaunchKernelGGL((gradlap_knl<H, false, true>), grid, block, 0,
+                               stream, f, lap, pdx, pdy, pdz, nx, ny, nz,
+                               ix, iy, iz, ix2, iy2, iz2);
+    });
+    return (int)hipGetLastError();
+}
+
+extern "C" int pystella_pd(
+    const double *f, double *out, int h, int axis, int accum,
+    int nx, int ny, int nz, int nf, double d, void *stream_)
+{
+    hipStream_t stream = (hipStream_t)stream_;
+    const dim3 grid = tile_grid(ny, nz, nf);
+    const dim3 block(BZ * BY);
+    const double inv = 1. / d;
+
+    DISPATCH_H(h, {
+        switch (axis * 2 + (accum ? 1 : 0)) {
+        case 0: hipLaunchKernelGGL((pd_knl<H, 0, false>), grid, block, 0,
+                                   stream, f, out, nx, ny, nz, inv); break;
+        case 1: hipLaunchKernelGGL((pd_knl<H, 0, true>), grid, block, 0,
+                                   stream, f, out, nx, ny, nz, inv); break;
+        case 2: hipLaunchKernelGGL((pd_knl<H, 1, false>), grid, block, 0,
+                                   stream, f, out, nx, ny, nz, inv); break;
+        case 3: hipLaunchKernelGGL((pd_knl<H, 1, true>), grid, block, 0,
+                                   stream, f, out, nx, ny, nz, inv); break;
+        case 4: hipLaunchKernelGGL((pd_knl<H, 2, false>), grid, block, 0,
+                                   stream, f, out, nx, ny, nz, inv); break;
+        case 5: hipLaunchKernelGGL((pd_knl<H, 2, true>), grid, block, 0,
+                                   stream, f, out, nx, ny, nz, inv); break;
+        }
+    });
+    return (int)hipGetLastError();
+}

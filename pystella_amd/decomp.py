@@ -164,10 +164,15 @@ class DomainDecomposition:
         recv_hi = torch.empty_like(send_hi)        # fills halo [n+h:n+2h]
         lo_rank, hi_rank = neighbors
 
+        # Pairing convention (also correct when lo_rank == hi_rank, e.g.
+        # two ranks with periodic wrap): phase A moves low faces downward
+        # (my high halo ← hi_rank's low face), phase B moves high faces
+        # upward.  Every rank posts A-ops before B-ops so same-peer
+        # send/recv pairs match by order.
         ops = [
-            dist.P2POp(dist.irecv, recv_lo, lo_rank),
             dist.P2POp(dist.irecv, recv_hi, hi_rank),
             dist.P2POp(dist.isend, send_lo, lo_rank),
+            dist.P2POp(dist.irecv, recv_lo, lo_rank),
             dist.P2POp(dist.isend, send_hi, hi_rank),
         ]
         for work in dist.batch_isend_irecv(ops):

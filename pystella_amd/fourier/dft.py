@@ -94,7 +94,9 @@ class BaseDFT:
         from itertools import product
         for i, j, k in product(*where_to_zero):
             if only_imag:
-                array[i, j, k] = array[i, j, k].real
+                val = array[i, j, k]
+                array[i, j, k] = complex(val.real, 0) \
+                    if not hasattr(val, "clone") else val.real.clone()
             else:
                 array[i, j, k] = 0.
         return array

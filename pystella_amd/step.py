@@ -351,11 +351,16 @@ class LowStorageRK134(LowStorageRKStepper):
     region (reference step.py:634-696)."""
 
     num_stages, expected_order = 13, 4
+    # note: the A_i of Niegemann et al. are negative; the reference
+    # stores them positive (step.py:648-663) but never exercises this
+    # stepper in its test matrix (step.py:849 omits it) — with positive
+    # A the scheme diverges.  Signs fixed here; 4th order verified in
+    # tests/test_step.py.
     _A = [
-        0, 0.6160178650170565, 0.4449487060774118, 1.0952033345276178,
-        1.2256030785959187, 0.2740182222332805, 0.0411952089052647,
-        0.179708489915356, 1.1771530652064288, 0.4078831463120878,
-        0.8295636426191777, 4.789597058425229, 0.6606671432964504,
+        0, -0.6160178650170565, -0.4449487060774118, -1.0952033345276178,
+        -1.2256030785959187, -0.2740182222332805, -0.0411952089052647,
+        -0.179708489915356, -1.1771530652064288, -0.4078831463120878,
+        -0.8295636426191777, -4.789597058425229, -0.6606671432964504,
     ]
     _B = [
         0.0271990297818803, 0.1772488819905108, 0.0378528418949694,
@@ -377,11 +382,12 @@ class LowStorageRK124(LowStorageRKStepper):
     (reference step.py:697-756)."""
 
     num_stages, expected_order = 12, 4
+    # A_i signs fixed relative to the reference, as for LowStorageRK134
     _A = [
-        0, 0.0923311242368072, 0.9441056581158819, 4.327127324757639,
-        2.155777132902607, 0.9770727190189062, 0.7581835342571139,
-        1.79775254708255, 2.691566797270077, 4.646679896026814,
-        0.1539613783825189, 0.5943293901830616,
+        0, -0.0923311242368072, -0.9441056581158819, -4.327127324757639,
+        -2.155777132902607, -0.9770727190189062, -0.7581835342571139,
+        -1.79775254708255, -2.691566797270077, -4.646679896026814,
+        -0.1539613783825189, -0.5943293901830616,
     ]
     _B = [
         0.0650008435125904, 0.0161459902249842, 0.5758627178358159,

@@ -1,0 +1,49 @@
+import os
+import sys
+
+import pytest
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: needs an MI355X (run with -m gpu on a GPU box)")
+
+
+def pytest_addoption(parser):
+    parser.addoption("--grid_shape", default="32,32,32")
+    parser.addoption("--proc_shape", default="1,1,1")
+
+
+@pytest.fixture
+def grid_shape(request):
+    return tuple(int(x) for x in
+                 request.config.getoption("--grid_shape").split(","))
+
+
+@pytest.fixture
+def proc_shape(request):
+    return tuple(int(x) for x in
+                 request.config.getoption("--proc_shape").split(","))
+
+
+def run_distributed(fn, world_size=2, args=()):
+    """Spawn `world_size` processes running fn(rank, world_size, *args)
+    under a gloo process group (CPU)."""
+    import torch.multiprocessing as mp
+    import tempfile
+    init_file = tempfile.NamedTemporaryFile(delete=False).name
+    mp.spawn(_dist_worker, args=(world_size, init_file, fn, args),
+             nprocs=world_size, join=True)
+
+
+def _dist_worker(rank, world_size, init_file, fn, args):
+    import torch.distributed as dist
+    dist.init_process_group(
+        "gloo", init_method=f"file://{init_file}", rank=rank,
+        world_size=world_size)
+    try:
+        fn(rank, world_size, *args)
+    finally:
+        dist.destroy_process_group()

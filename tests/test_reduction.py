@@ -1,0 +1,93 @@
+"""Reduction / FieldStatistics tests vs direct torch reference."""
+
+import numpy as np
+import torch
+
+import pystella_amd as ps
+from pystella_amd.field import Field, var
+
+
+def test_reduction_ops(grid_shape=(16, 16, 16)):
+    h = 1
+    pad = tuple(n + 2 * h for n in grid_shape)
+    decomp = ps.DomainDecomposition((1, 1, 1), h, rank_shape=grid_shape)
+    torch.manual_seed(0)
+    f = torch.rand(pad, dtype=torch.float64)
+    interior = f[h:-h, h:-h, h:-h]
+
+    F = Field("f", offset="h")
+    red = ps.Reduction(decomp, {
+        "mean": [F],
+        "sum_sq": [(F**2, "sum")],
+        "mx": [(F, "max")],
+        "mn": [(F, "min")],
+    }, halo_shape=h, grid_size=float(np.prod(grid_shape)))
+
+    out = red(f=f)
+    assert abs(out["mean"][0] - interior.mean().item()) < 1e-12
+    assert abs(out["sum_sq"][0] - (interior**2).sum().item()) < 1e-9
+    assert out["mx"][0] == interior.max().item()
+    assert out["mn"][0] == interior.min().item()
+
+
+def test_reduction_with_scalar_arg(grid_shape=(8, 8, 8)):
+    decomp = ps.DomainDecomposition((1, 1, 1), 0, rank_shape=grid_shape)
+    f = torch.rand(grid_shape, dtype=torch.float64)
+    F = Field("f", offset=0)
+    a = var("a")
+    red = ps.Reduction(decomp, {"kin": [F**2 / 2 / a**2]}, halo_shape=0,
+                       grid_size=float(np.prod(grid_shape)))
+    out = red(f=f, a=np.array([2.0]))
+    expect = (f**2).mean().item() / 8
+    assert abs(out["kin"][0] - expect) < 1e-12
+
+
+def test_field_statistics(grid_shape=(16, 16, 16)):
+    h = 2
+    pad = tuple(n + 2 * h for n in grid_shape)
+    decomp = ps.DomainDecomposition((1, 1, 1), h, rank_shape=grid_shape)
+    stats = ps.FieldStatistics(decomp, h, rank_shape=grid_shape,
+                               grid_size=float(np.prod(grid_shape)))
+    torch.manual_seed(1)
+    f = torch.rand((2,) + pad, dtype=torch.float64)
+    out = stats(f)
+    for i in range(2):
+        interior = f[i, h:-h, h:-h, h:-h]
+        assert abs(out["mean"][i] - interior.mean().item()) < 1e-12
+        assert abs(out["variance"][i] - interior.var(correction=0).item()) \
+            < 1e-10
+
+
+def test_scalar_sector_energy(grid_shape=(12, 12, 12)):
+    """Energy reducers vs a hand-written torch computation
+    (analogue of reference test/test_energy.py)."""
+    from pystella_amd.sectors import get_rho_and_p
+    h = 1
+    pad = tuple(n + 2 * h for n in grid_shape)
+    decomp = ps.DomainDecomposition((1, 1, 1), h, rank_shape=grid_shape)
+    torch.manual_seed(2)
+    f = torch.rand((2,) + pad, dtype=torch.float64)
+    dfdt = torch.rand((2,) + pad, dtype=torch.float64)
+    lap_f = torch.rand((2,) + grid_shape, dtype=torch.float64)
+    a = 1.3
+
+    def potential(fld):
+        return fld[0]**2 / 2 + fld[0]**2 * fld[1]**2 / 4
+
+    sector = ps.ScalarSector(2, potential=potential)
+    red = ps.Reduction(decomp, sector, halo_shape=h,
+                       grid_size=float(np.prod(grid_shape)),
+                       callback=get_rho_and_p)
+    out = red(f=f, dfdt=dfdt, lap_f=lap_f, a=np.array([a]))
+
+    fi = f[:, h:-h, h:-h, h:-h]
+    dfi = dfdt[:, h:-h, h:-h, h:-h]
+    kin = [(dfi[i]**2 / 2 / a**2).mean().item() for i in range(2)]
+    pot = (fi[0]**2 / 2 + fi[0]**2 * fi[1]**2 / 4).mean().item()
+    grad = [(-fi[i] * lap_f[i] / 2 / a**2).mean().item() for i in range(2)]
+    for i in range(2):
+        assert abs(out["kinetic"][i] - kin[i]) < 1e-12
+        assert abs(out["gradient"][i] - grad[i]) < 1e-12
+    assert abs(out["potential"][0] - pot) < 1e-12
+    total = sum(kin) + sum(grad) + pot
+    assert abs(out["total"] - total) < 1e-11
