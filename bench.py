@@ -77,8 +77,9 @@ def main():
     stepper = ps.LowStorageRK54([sector], halo_shape=h,
                                 rank_shape=rank_shape, dt=dt)
     derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=rank_shape)
-    reduce_energy = ps.Reduction(
-        decomp, sector, halo_shape=h, callback=get_rho_and_p,
+    from pystella_amd.fusion import FusedLaplacianReduction
+    reduce_energy = FusedLaplacianReduction(
+        decomp, sector, derivs, halo_shape=h, callback=get_rho_and_p,
         rank_shape=rank_shape, grid_size=grid_size)
 
     gen = torch.Generator(device="cpu").manual_seed(7 + decomp.rank)
@@ -93,7 +94,7 @@ def main():
     energy = None
 
     def compute_energy(a):
-        derivs(fx=f, lap=lap_f)
+        # fused: halo exchange + Laplacian stencil + energy reduction
         return reduce_energy(f=f, dfdt=dfdt, lap_f=lap_f, a=np.array(a))
 
     energy = compute_energy(1.)

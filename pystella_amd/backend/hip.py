@@ -80,14 +80,10 @@ ELEMENTWISE_TEMPLATE = """{defines}
 extern "C" __global__ __launch_bounds__(256) void {name}(
     {params})
 {{
-    long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
-    const long total = (long)NX * NY * NZ;
-    const long stride = (long)gridDim.x * blockDim.x;
-    for (; idx < total; idx += stride) {{
-        const int k = (int)(idx % NZ);
-        const long t = idx / NZ;
-        const int j = (int)(t % NY);
-        const int i = (int)(t / NY);
+    const int k = blockIdx.x * 64 + (threadIdx.x % 64);
+    const int j = blockIdx.y * 4 + (threadIdx.x / 64);
+    if (k >= NZ || j >= NY) return;
+    for (int i = 0; i < NX; ++i) {{
         {body}
     }}
 }}
@@ -114,8 +110,8 @@ class JitElementwise:
         self.scalar_keys = [k for _, k in cg.scalars]
         self.key = ext().jit_compile(src, name)
 
-        total = int(np.prod(rank_shape))
-        self.grid = min((total + 255) // 256, 4096)
+        nx, ny, nz = rank_shape
+        self.grid = ((nz + 63) // 64, (ny + 3) // 4)
 
     def __call__(self, env):
         ptrs = []
@@ -123,8 +119,8 @@ class JitElementwise:
             t = _check_tensor(fa.name, env[fa.name])
             ptrs.append(t.data_ptr())
         doubles = [_resolve_scalar(env, k) for k in self.scalar_keys]
-        ext().jit_launch(self.key, self.grid, 1, 1, 256, 1, 1, 0,
-                         _stream(), ptrs, [], doubles)
+        ext().jit_launch(self.key, self.grid[0], self.grid[1], 1,
+                         256, 1, 1, 0, _stream(), ptrs, [], doubles)
 
 
 def get_elementwise_kernel(map_dict, tmp_instructions, field_args,
@@ -144,17 +140,16 @@ extern "C" __global__ __launch_bounds__(256) void {name}(
 {{
     double acc[NRED];
     {init}
-    long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
-    const long total = (long)NX * NY * NZ;
-    const long stride = (long)gridDim.x * blockDim.x;
-    for (; idx < total; idx += stride) {{
-        const int k = (int)(idx % NZ);
-        const long t = idx / NZ;
-        const int j = (int)(t % NY);
-        const int i = (int)(t / NY);
-        {body}
+    const int k = blockIdx.x * 64 + (threadIdx.x % 64);
+    const int j = blockIdx.y * 4 + (threadIdx.x / 64);
+    if (k < NZ && j < NY) {{
+        for (int i = 0; i < NX; ++i) {{
+            {body}
+        }}
     }}
     __shared__ double sd[256];
+    const int nblk = gridDim.x * gridDim.y;
+    const int bid = blockIdx.y * gridDim.x + blockIdx.x;
     for (int r = 0; r < NRED; ++r) {{
         sd[threadIdx.x] = acc[r];
         __syncthreads();
@@ -165,7 +160,7 @@ extern "C" __global__ __launch_bounds__(256) void {name}(
             __syncthreads();
         }}
         if (threadIdx.x == 0)
-            partials[(long)r * gridDim.x + blockIdx.x] = sd[0];
+            partials[(long)r * nblk + bid] = sd[0];
         __syncthreads();
     }}
 }}
@@ -220,9 +215,26 @@ class JitReduction:
         self.scalar_keys = [k for _, k in cg.scalars]
         self.key = ext().jit_compile(src, name)
 
-        total = int(np.prod(rank_shape))
-        self.grid = min((total + 255) // 256, 2048)
+        nx, ny, nz = rank_shape
+        self.grid = ((nz + 63) // 64, (ny + 3) // 4)
+        self.nblk = self.grid[0] * self.grid[1]
         self._partials = None
+
+    def _finish(self, dev):
+        """Combine per-block partials on-device, one packed D2H copy."""
+        vals = torch.empty(len(self.entries), dtype=torch.float64,
+                           device=dev)
+        for r, (_, op) in enumerate(self.entries):
+            row = self._partials[r]
+            if op in ("sum", "avg"):
+                vals[r] = row.sum()
+            elif op == "prod":
+                vals[r] = row.prod()
+            elif op == "max":
+                vals[r] = row.max()
+            else:
+                vals[r] = row.min()
+        return vals.cpu().tolist()
 
     def __call__(self, env):
         dev = None
@@ -234,25 +246,14 @@ class JitReduction:
         nred = len(self.entries)
         if (self._partials is None
                 or self._partials.device != dev
-                or self._partials.shape[1] != self.grid):
-            self._partials = torch.empty((nred, self.grid),
+                or self._partials.shape[1] != self.nblk):
+            self._partials = torch.empty((nred, self.nblk),
                                          dtype=torch.float64, device=dev)
         doubles = [_resolve_scalar(env, k) for k in self.scalar_keys]
-        ext().jit_launch(self.key, self.grid, 1, 1, 256, 1, 1, 0,
-                         _stream(), ptrs + [self._partials.data_ptr()],
-                         [], doubles)
-        out = []
-        for r, (_, op) in enumerate(self.entries):
-            row = self._partials[r]
-            if op in ("sum", "avg"):
-                out.append(row.sum().item())
-            elif op == "prod":
-                out.append(row.prod().item())
-            elif op == "max":
-                out.append(row.max().item())
-            else:
-                out.append(row.min().item())
-        return out
+        ext().jit_launch(self.key, self.grid[0], self.grid[1], 1,
+                         256, 1, 1, 0, _stream(),
+                         ptrs + [self._partials.data_ptr()], [], doubles)
+        return self._finish(dev)
 
 
 def get_reduction_kernel(entries, field_args, scalar_names, halo,
@@ -356,14 +357,15 @@ def _flat_fields(t, ndim_grid=3):
     return t, nf
 
 
-def derivs(fx, lap=None, pdx=None, pdy=None, pdz=None, halo=None, dx=None,
-           h=None, stream=True):
+def derivs(fx, lap=None, pdx=None, pdy=None, pdz=None, grd=None, halo=None,
+           dx=None, h=None, stream=True):
     if len(set(halo)) != 1:
         raise NotImplementedError("GPU stencils require isotropic halo")
     _check_tensor("fx", fx)
     nxp, nyp, nzp = fx.shape[-3:]
     nx, ny, nz = nxp - 2 * h, nyp - 2 * h, nzp - 2 * h
     _, nf = _flat_fields(fx)
+    uvol = nx * ny * nz
 
     def ptr(t):
         if t is None:
@@ -372,18 +374,26 @@ def derivs(fx, lap=None, pdx=None, pdy=None, pdz=None, halo=None, dx=None,
         return t.data_ptr()
 
     e = ext()
-    want_grad = pdx is not None and pdy is not None and pdz is not None
+    # gradient outputs: either a packed (..., 3, nx, ny, nz) grd array
+    # (per-field component stride 3*uvol) or three standalone arrays
+    if grd is not None and isinstance(grd, torch.Tensor):
+        _check_tensor("grd", grd)
+        gp = grd.data_ptr()
+        px, py, pz_ = gp, gp + 8 * uvol, gp + 16 * uvol
+        g_fstride = 3 * uvol
+        want_grad = True
+    elif pdx is not None and pdy is not None and pdz is not None:
+        px, py, pz_ = ptr(pdx), ptr(pdy), ptr(pdz)
+        g_fstride = uvol
+        want_grad = True
+    else:
+        px = py = pz_ = 0
+        g_fstride = 0
+        want_grad = False
+
     if lap is not None or want_grad:
-        if lap is not None and not want_grad:
-            e.gradlap(fx.data_ptr(), ptr(lap), 0, 0, 0, h, nx, ny, nz, nf,
-                      dx[0], dx[1], dx[2], _stream())
-        elif want_grad and lap is None:
-            e.gradlap(fx.data_ptr(), 0, ptr(pdx), ptr(pdy), ptr(pdz), h,
-                      nx, ny, nz, nf, dx[0], dx[1], dx[2], _stream())
-        else:
-            e.gradlap(fx.data_ptr(), ptr(lap), ptr(pdx), ptr(pdy),
-                      ptr(pdz), h, nx, ny, nz, nf, dx[0], dx[1], dx[2],
-                      _stream())
+        e.gradlap(fx.data_ptr(), ptr(lap), px, py, pz_, g_fstride,
+                  h, nx, ny, nz, nf, dx[0], dx[1], dx[2], _stream())
         return
     # single-axis derivatives
     for axis, out in enumerate((pdx, pdy, pdz)):
@@ -407,3 +417,192 @@ def divergence(vec, div, halo=None, dx=None, h=None):
              nx, ny, nz, 1, dx[1], _stream())
         e.pd(vec[s][2].data_ptr(), div[s].data_ptr(), h, 2, 1,
              nx, ny, nz, 1, dx[2], _stream())
+
+
+# ---------------------------------------------------------------------------
+# Fused Laplacian + reduction kernel (MI355X-specific optimization).
+#
+# The reference hot loop runs the stencil pass and the energy reduction
+# as separate kernels (reference examples/scalar_preheating.py:258-271 →
+# derivs.py:339-429 then reduction.py:206); that re-reads f and lap_f
+# from HBM.  Here one x-marching pass computes lap (register ring +
+# current-plane neighbor loads), stores it, and accumulates the energy
+# reductions with the freshly computed Laplacian still in registers:
+# HBM traffic per site drops from (f, lap w, f, dfdt, lap r) to
+# (f, dfdt, lap w).
+
+LAPRED_TEMPLATE = """{defines}
+{preamble}
+#define NRED {nred}
+#define NF {nf}
+extern "C" __global__ __launch_bounds__(256) void {name}(
+    {params})
+{{
+    double acc[NRED];
+    {init}
+    const int k = blockIdx.x * 64 + (threadIdx.x % 64);
+    const int j = blockIdx.y * 4 + (threadIdx.x / 64);
+    if (k < NZ && j < NY) {{
+        const long sx = PSY * PSZ;
+        double ring[NF][2 * H + 1];
+        #pragma unroll
+        for (int fld = 0; fld < NF; ++fld) {{
+            const double* fp = {fname} + (long)fld * PVOL
+                               + (long)(j + H) * PSZ + (k + H);
+            #pragma unroll
+            for (int p = 0; p < 2 * H; ++p)
+                ring[fld][p] = fp[(long)p * sx];
+        }}
+        for (int i = 0; i < NX; ++i) {{
+            double lapv[NF];
+            #pragma unroll
+            for (int fld = 0; fld < NF; ++fld) {{
+                const double* fp = {fname} + (long)fld * PVOL
+                                   + (long)(j + H) * PSZ + (k + H);
+                ring[fld][2 * H] = fp[(long)(i + 2 * H) * sx];
+                const double* cp = fp + (long)(i + H) * sx;
+                double la = ring[fld][H] * LAPC0;
+                {lap_terms}
+                lapv[fld] = la;
+                {lapname}[(long)fld * UVOL + (((long)i * NY + j) * NZ + k)]
+                    = la;
+            }}
+            {body}
+            #pragma unroll
+            for (int fld = 0; fld < NF; ++fld)
+                #pragma unroll
+                for (int p = 0; p < 2 * H; ++p)
+                    ring[fld][p] = ring[fld][p + 1];
+        }}
+    }}
+    __shared__ double sd[256];
+    const int nblk = gridDim.x * gridDim.y;
+    const int bid = blockIdx.y * gridDim.x + blockIdx.x;
+    for (int r = 0; r < NRED; ++r) {{
+        sd[threadIdx.x] = acc[r];
+        __syncthreads();
+        for (int s = 128; s > 0; s >>= 1) {{
+            if (threadIdx.x < s)
+                sd[threadIdx.x] = COMBINE(r, sd[threadIdx.x],
+                                          sd[threadIdx.x + s]);
+            __syncthreads();
+        }}
+        if (threadIdx.x == 0)
+            partials[(long)r * nblk + bid] = sd[0];
+        __syncthreads();
+    }}
+}}
+"""
+
+
+class _LapCodegen(Codegen):
+    """Codegen that maps accesses to the stencil field's center value and
+    the freshly computed Laplacian onto kernel registers."""
+
+    def __init__(self, field_args, halo, rank_shape, f_name, lap_name):
+        super().__init__(field_args, halo, rank_shape)
+        self.f_name = f_name
+        self.lap_name = lap_name
+
+    def field_access(self, f, outer_idx):
+        if f.is_spatial and not any(f.shift) and len(outer_idx) <= 1:
+            lin = int(outer_idx[0]) if outer_idx else 0
+            if f.name == self.lap_name:
+                return f"lapv[{lin}]"
+            if f.name == self.f_name:
+                return f"ring[{lin}][H]"
+        return super().field_access(f, outer_idx)
+
+
+class JitLapReduction:
+    """Fused lap-stencil + multi-quantity reduction (see module note)."""
+
+    def __init__(self, entries, field_args, scalar_names, halo, rank_shape,
+                 dx, nf, f_name="f", lap_name="lap_f", name="lapred_map"):
+        from pystella_amd.derivs import _LAP_COEFS
+        self.rank_shape = tuple(rank_shape)
+        self.entries = entries
+        h = max(halo) if isinstance(halo, (tuple, list)) else halo
+        self.nf = nf
+        cg = _LapCodegen(field_args, halo, rank_shape, f_name, lap_name)
+
+        init_lines, body_lines, combine_cases = [], [], []
+        for r, (expr, op) in enumerate(entries):
+            init_lines.append(f"acc[{r}] = {_OP_INIT[op]};")
+            comb = _OP_COMBINE[op]
+            val = cg.emit(expr)
+            body_lines.append(
+                "{ const double a = acc[%d]; const double b = %s; "
+                "acc[%d] = %s; }" % (r, val, r, comb))
+            combine_cases.append(f"(r == {r}) ? {comb} : ")
+        combine = "".join(combine_cases) + "0.0"
+
+        # Laplacian stencil terms with dx baked in
+        inv2 = [1.0 / d / d for d in dx]
+        coefs = _LAP_COEFS[h]
+        lap_terms = []
+        for s in range(1, h + 1):
+            c = coefs[s]
+            lap_terms.append(
+                f"la += {c!r} * ((ring[fld][H+{s}] + ring[fld][H-{s}])"
+                f"*{inv2[0]!r} + (cp[{s}*PSZ] + cp[-{s}*PSZ])*{inv2[1]!r}"
+                f" + (cp[{s}] + cp[-{s}])*{inv2[2]!r});")
+        lapc0 = coefs[0] * (inv2[0] + inv2[1] + inv2[2])
+
+        # pointer params: stencil field first, then lap, then the rest
+        self.ptr_names = [f_name, lap_name] + sorted(
+            fa.name for fa in field_args
+            if fa.spatial and fa.name not in (f_name, lap_name))
+        by_name = {fa.name: fa for fa in field_args}
+        self.field_args = [by_name[n] for n in self.ptr_names if n in by_name]
+        ptr_params = ", ".join(
+            f"double* __restrict__ {n}" for n in self.ptr_names)
+        dbl_params = ", ".join(f"double {c}" for c, _ in cg.scalars)
+        params = ", ".join(x for x in (
+            ptr_params, "double* __restrict__ partials", dbl_params) if x)
+
+        defines = geometry_defines(halo, rank_shape)
+        defines += f"#define COMBINE(r, a, b) ({combine})\n"
+        defines += f"#define LAPC0 ({lapc0!r})\n"
+        src = LAPRED_TEMPLATE.format(
+            defines=defines, preamble=PREAMBLE, nred=len(entries), nf=nf,
+            name=name, params=params, fname=f_name, lapname=lap_name,
+            init="\n    ".join(init_lines),
+            lap_terms="\n                ".join(lap_terms),
+            body="\n            ".join(body_lines))
+        self.source = src
+        self.scalar_keys = [k for _, k in cg.scalars]
+        self.key = ext().jit_compile(src, name)
+
+        nx, ny, nz = rank_shape
+        self.grid = ((nz + 63) // 64, (ny + 3) // 4)
+        self.nblk = self.grid[0] * self.grid[1]
+        self._partials = None
+
+    _finish = JitReduction._finish
+
+    def __call__(self, env):
+        dev = None
+        ptrs = []
+        for n in self.ptr_names:
+            t = _check_tensor(n, env[n])
+            dev = t.device
+            ptrs.append(t.data_ptr())
+        nred = len(self.entries)
+        if (self._partials is None
+                or self._partials.device != dev
+                or self._partials.shape[1] != self.nblk):
+            self._partials = torch.empty((nred, self.nblk),
+                                         dtype=torch.float64, device=dev)
+        doubles = [_resolve_scalar(env, k) for k in self.scalar_keys]
+        ext().jit_launch(self.key, self.grid[0], self.grid[1], 1,
+                         256, 1, 1, 0, _stream(),
+                         ptrs + [self._partials.data_ptr()], [], doubles)
+        return self._finish(dev)
+
+
+def get_lap_reduction_kernel(entries, field_args, scalar_names, halo,
+                             rank_shape, dx, nf, f_name="f",
+                             lap_name="lap_f"):
+    return JitLapReduction(entries, field_args, scalar_names, halo,
+                           rank_shape, dx, nf, f_name, lap_name)

@@ -74,11 +74,14 @@ constexpr int BY = 4;
 
 // Fused gradient/Laplacian kernel, x-marching with a register ring.
 // f is halo-padded (nx+2H, ny+2H, nz+2H) per field; outputs unpadded.
+// Gradient outputs may live inside a packed (nf, 3, nx, ny, nz) "grd"
+// array: g_fstride is the per-field stride of each gradient component
+// (3*uvol when packed, uvol for standalone component arrays).
 template <int H, bool LAP, bool GRAD>
 __global__ __launch_bounds__(BZ * BY) void gradlap_knl(
     const double *__restrict__ f, double *__restrict__ lap,
     double *__restrict__ pdx, double *__restrict__ pdy,
-    double *__restrict__ pdz, int nx, int ny, int nz,
+    double *__restrict__ pdz, int64_t g_fstride, int nx, int ny, int nz,
     double inv_dx, double inv_dy, double inv_dz,
     double inv_dx2, double inv_dy2, double inv_dz2)
 {
@@ -98,11 +101,11 @@ __global__ __launch_bounds__(BZ * BY) void gradlap_knl(
                        + (k + H);
     double *outl = LAP ? lap + (int64_t)fld * uvol + (int64_t)j * nz + k
                        : nullptr;
-    double *outx = GRAD ? pdx + (int64_t)fld * uvol + (int64_t)j * nz + k
+    double *outx = GRAD ? pdx + fld * g_fstride + (int64_t)j * nz + k
                         : nullptr;
-    double *outy = GRAD ? pdy + (int64_t)fld * uvol + (int64_t)j * nz + k
+    double *outy = GRAD ? pdy + fld * g_fstride + (int64_t)j * nz + k
                         : nullptr;
-    double *outz = GRAD ? pdz + (int64_t)fld * uvol + (int64_t)j * nz + k
+    double *outz = GRAD ? pdz + fld * g_fstride + (int64_t)j * nz + k
                         : nullptr;
     const int64_t so = (int64_t)ny * nz;          // unpadded x stride
 
@@ -204,7 +207,7 @@ inline dim3 tile_grid(int ny, int nz, int nf)
 
 extern "C" int pystella_gradlap(
     const double *f, double *lap, double *pdx, double *pdy, double *pdz,
-    int h, int nx, int ny, int nz, int nf,
+    long long g_fstride, int h, int nx, int ny, int nz, int nf,
     double dx, double dy, double dz, void *stream_)
 {
     hipStream_t stream = (hipStream_t)stream_;
@@ -218,15 +221,18 @@ extern "C" int pystella_gradlap(
     DISPATCH_H(h, {
         if (do_lap && do_grad)
             hipLaunchKernelGGL((gradlap_knl<H, true, true>), grid, block, 0,
-                               stream, f, lap, pdx, pdy, pdz, nx, ny, nz,
+                               stream, f, lap, pdx, pdy, pdz,
+                               (int64_t)g_fstride, nx, ny, nz,
                                ix, iy, iz, ix2, iy2, iz2);
         else if (do_lap)
             hipLaunchKernelGGL((gradlap_knl<H, true, false>), grid, block, 0,
-                               stream, f, lap, pdx, pdy, pdz, nx, ny, nz,
+                               stream, f, lap, pdx, pdy, pdz,
+                               (int64_t)g_fstride, nx, ny, nz,
                                ix, iy, iz, ix2, iy2, iz2);
         else
             hipLaunchKernelGGL((gradlap_knl<H, false, true>), grid, block, 0,
-                               stream, f, lap, pdx, pdy, pdz, nx, ny, nz,
+                               stream, f, lap, pdx, pdy, pdz,
+                               (int64_t)g_fstride, nx, ny, nz,
                                ix, iy, iz, ix2, iy2, iz2);
     });
     return (int)hipGetLastError();

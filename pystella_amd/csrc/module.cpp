@@ -41,8 +41,8 @@ namespace py = pybind11;
 // ---------------------------------------------------------------------------
 // AOT kernels (derivs.hip)
 extern "C" int pystella_gradlap(const double *, double *, double *, double *,
-                                double *, int, int, int, int, int, double,
-                                double, double, void *);
+                                double *, long long, int, int, int, int, int,
+                                double, double, double, void *);
 extern "C" int pystella_pd(const double *, double *, int, int, int, int, int,
                            int, int, double, void *);
 
@@ -54,13 +54,14 @@ static void check_knl(int err, const char *what)
 }
 
 static void gradlap(uintptr_t f, uintptr_t lap, uintptr_t pdx, uintptr_t pdy,
-                    uintptr_t pdz, int h, int nx, int ny, int nz, int nf,
-                    double dx, double dy, double dz, uintptr_t stream)
+                    uintptr_t pdz, int64_t g_fstride, int h, int nx, int ny,
+                    int nz, int nf, double dx, double dy, double dz,
+                    uintptr_t stream)
 {
     check_knl(pystella_gradlap((const double *)f, (double *)lap,
                                (double *)pdx, (double *)pdy, (double *)pdz,
-                               h, nx, ny, nz, nf, dx, dy, dz,
-                               (void *)stream),
+                               (long long)g_fstride, h, nx, ny, nz, nf,
+                               dx, dy, dz, (void *)stream),
               "gradlap");
 }
 

@@ -195,10 +195,12 @@ class FiniteDifferencer:
             # allow positional call derivs(fx, ...)
             fx = queue
             queue = None
+        grd_tensor = None
         if grd is not None:
             if isinstance(grd, (tuple, list)):
                 pdx, pdy, pdz = grd
             else:
+                grd_tensor = grd
                 pdx = grd[..., 0, :, :, :]
                 pdy = grd[..., 1, :, :, :]
                 pdz = grd[..., 2, :, :, :]
@@ -206,7 +208,7 @@ class FiniteDifferencer:
         self.decomp.share_halos(fx)
 
         if fx.is_cuda:
-            self._call_hip(fx, lap, pdx, pdy, pdz)
+            self._call_hip(fx, lap, pdx, pdy, pdz, grd_tensor)
             return
 
         from itertools import product
@@ -243,11 +245,15 @@ class FiniteDifferencer:
             self._hip = hip
         return self._hip
 
-    def _call_hip(self, fx, lap, pdx, pdy, pdz):
+    def _call_hip(self, fx, lap, pdx, pdy, pdz, grd_tensor=None):
         hip = self._hip_mod()
-        hip.derivs(fx, lap=lap, pdx=pdx, pdy=pdy, pdz=pdz,
-                   halo=self.halo_shape, dx=self.dx, h=self._h,
-                   stream=self.stream)
+        if grd_tensor is not None:
+            hip.derivs(fx, lap=lap, grd=grd_tensor, halo=self.halo_shape,
+                       dx=self.dx, h=self._h, stream=self.stream)
+        else:
+            hip.derivs(fx, lap=lap, pdx=pdx, pdy=pdy, pdz=pdz,
+                       halo=self.halo_shape, dx=self.dx, h=self._h,
+                       stream=self.stream)
 
     def _call_hip_div(self, vec, div):
         hip = self._hip_mod()

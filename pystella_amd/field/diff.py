@@ -39,8 +39,9 @@ def _diff(expr, x):
         return 0
     if _leaves_equal(expr, x):
         return 1
-    if isinstance(expr, (Variable, Subscript)):
-        # distinct leaf (includes Field)
+    from pystella_amd.field import Field
+    if isinstance(expr, (Variable, Subscript, Field)):
+        # distinct leaf
         return 0
     if isinstance(expr, Sum):
         return flattened_sum(tuple(_diff(c, x) for c in expr.children

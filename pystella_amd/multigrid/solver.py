@@ -1,16 +1,251 @@
-"""Multigrid drivers: Full Approximation Scheme and linear MG.
+"""Multigrid drivers: Full Approximation Scheme (FAS) and linear MG.
 
-Analogue of reference pystella/multigrid/__init__.py:169-493.
-
-Status: full implementation arrives with the multigrid milestone.
+Analogue of reference pystella/multigrid/__init__.py:169-493.  Each
+level gets its own :class:`~pystella_amd.DomainDecomposition`; arrays
+are torch tensors on whatever device the fine-level arrays live on.
 """
+
+from __future__ import annotations
+
+import numpy as np
+import torch
+
+from pystella_amd.multigrid.transfer import (
+    FullWeighting, LinearInterpolation,
+)
+
+__all__ = ["FullApproximationScheme", "MultiGridSolver"]
+
+
+def mu_cycle(mu, i, nu1, nu2, max_depth):
+    """Generic μ-cycle as (level, iterations) events
+    (reference multigrid/__init__.py:55-81)."""
+    if i == max_depth:
+        return [(i, nu2)]
+    x = mu_cycle(mu, i + 1, nu1, nu2, max_depth)
+    return [(i, nu1)] + x + x[1:] * (mu - 1) + [(i, nu2)]
+
+
+def v_cycle(nu1, nu2, max_depth):
+    return mu_cycle(1, 0, nu1, nu2, max_depth)
+
+
+def w_cycle(nu1, nu2, max_depth):
+    return mu_cycle(2, 0, nu1, nu2, max_depth)
+
+
+def _updown(i, j, k, nu1, nu2):
+    down = [(a, nu1) for a in range(i, j)]
+    up = [(a, nu2) for a in range(j, k - 1, -1)]
+    return down + up
+
+
+def f_cycle(nu1, nu2, max_depth):
+    """F-cycle (reference multigrid/__init__.py:140-166)."""
+    cycle = _updown(0, max_depth, max_depth - 1, nu1, nu2)
+    for top in range(max_depth - 1, 0, -1):
+        cycle += _updown(top + 1, max_depth, top - 1, nu1, nu2)
+    return cycle
 
 
 class FullApproximationScheme:
-    def __init__(self, *a, **kw):
-        raise NotImplementedError("FAS multigrid: in progress")
+    """FAS driver for (possibly nonlinear) boundary-value problems
+    (reference multigrid/__init__.py:169-439).
+
+    :arg solver: a :class:`~pystella_amd.multigrid.relax.RelaxationBase`
+        subclass instance.
+    :arg halo_shape: halo layers (int).
+    """
+
+    def __init__(self, solver, halo_shape, **kwargs):
+        self.solver = solver
+        self.halo_shape = halo_shape
+
+        Restrictor = kwargs.pop("Restrictor", FullWeighting)
+        self.restrict = Restrictor(halo_shape=halo_shape)
+        self.restrict_and_correct = Restrictor(halo_shape=halo_shape,
+                                               correct=True)
+        Interpolator = kwargs.pop("Interpolator", LinearInterpolation)
+        self.interpolate = Interpolator(halo_shape=halo_shape)
+        self.interpolate_and_correct = Interpolator(halo_shape=halo_shape,
+                                                    correct=True)
+
+        self.unknowns = {}
+        self.rhos = {}
+        self.auxiliaries = {}
+        self.tmp = {}
+        self.resid = {}
+        self.dx = {}
+        self.decomp = {}
+        self.smooth_args = {}
+        self.resid_args = {}
+
+    # -- helpers ------------------------------------------------------------
+    def coarse_array_like(self, f1h):
+        h = self.halo_shape
+
+        def halve_and_pad(n):
+            return (n - 2 * h) // 2 + 2 * h
+
+        shape = tuple(map(halve_and_pad, f1h.shape))
+        return torch.zeros(shape, dtype=f1h.dtype, device=f1h.device)
+
+    def coarse_level_like(self, dict_1):
+        return {k: self.coarse_array_like(v) for k, v in dict_1.items()}
+
+    # -- level transfers ----------------------------------------------------
+    def transfer_down(self, queue=None, i=None):
+        """Fine level i−1 → coarse level i: restrict unknowns, restrict
+        the fine residual, and build the FAS coarse rhs
+        (reference multigrid/__init__.py:244-267)."""
+        if i is None:
+            i = queue
+        for key, f1 in self.unknowns[i - 1].items():
+            f2 = self.unknowns[i][key]
+            self.restrict(f1=f1, f2=f2)
+            self.decomp[i].share_halos(f2)
+
+        self.solver.residual(**self.resid_args[i - 1])
+
+        for key, r1 in self.resid[i - 1].items():
+            r2 = self.resid[i][key]
+            self.decomp[i - 1].share_halos(r1)
+            self.restrict(f1=r1, f2=r2)
+
+        self.solver.lhs_correction(**self.resid_args[i])
+        for rho in self.rhos[i].values():
+            self.decomp[i].share_halos(rho)
+
+    def transfer_up(self, queue=None, i=None):
+        """Coarse level i+1 → fine level i: FAS coarse-grid correction
+        (reference multigrid/__init__.py:269-283)."""
+        if i is None:
+            i = queue
+        for key, f1 in self.unknowns[i].items():
+            f2 = self.unknowns[i + 1][key]
+            self.restrict_and_correct(f1=f1, f2=f2)
+            self.decomp[i + 1].share_halos(f2)
+            self.interpolate_and_correct(f1=f1, f2=f2)
+            self.decomp[i].share_halos(f1)
+
+    def smooth(self, queue=None, i=None, nu=None):
+        if nu is None:
+            queue, i, nu = None, queue, i
+        errs1 = self.solver.get_error(**self.resid_args[i])
+        self.solver(self.decomp[i], iterations=nu, **self.smooth_args[i])
+        errs2 = self.solver.get_error(**self.resid_args[i])
+        return [(i, errs1), (i, errs2)]
+
+    # -- setup --------------------------------------------------------------
+    def setup(self, decomp0, queue=None, dx0=None, depth=None, **kwargs):
+        self.decomp[0] = decomp0
+        self.dx[0] = np.array(dx0)
+
+        self.unknowns[0] = {}
+        self.rhos[0] = {}
+        for k, v in self.solver.f_to_rho_dict.items():
+            self.unknowns[0][k] = kwargs.pop(k)
+            self.rhos[0][v] = kwargs.pop(v)
+        self.auxiliaries[0] = kwargs
+
+        if 0 not in self.tmp:
+            self.tmp[0] = {}
+            self.resid[0] = {}
+            for k, f in self.unknowns[0].items():
+                self.tmp[0]["tmp_" + k] = torch.zeros_like(f)
+                self.resid[0]["r_" + k] = self.tmp[0]["tmp_" + k]
+
+        from pystella_amd.decomp import DomainDecomposition
+        for i in range(depth + 1):
+            if i not in self.dx:
+                self.dx[i] = np.array(self.dx[i - 1] * 2)
+            if i not in self.decomp:
+                ng_2 = tuple(n // 2 for n in self.decomp[i - 1].rank_shape)
+                self.decomp[i] = DomainDecomposition(
+                    self.decomp[i - 1].proc_shape, self.halo_shape,
+                    rank_shape=ng_2)
+            if i not in self.unknowns:
+                self.unknowns[i] = self.coarse_level_like(
+                    self.unknowns[i - 1])
+            if i not in self.tmp:
+                self.tmp[i] = self.coarse_level_like(self.tmp[i - 1])
+                self.resid[i] = {}
+                for key in self.unknowns[i]:
+                    self.resid[i][f"r_{key}"] = self.tmp[i][f"tmp_{key}"]
+            if i not in self.rhos:
+                self.rhos[i] = self.coarse_level_like(self.rhos[i - 1])
+            if i not in self.auxiliaries:
+                self.auxiliaries[i] = self.coarse_level_like(
+                    self.auxiliaries[i - 1])
+                for k, f1 in self.auxiliaries[i - 1].items():
+                    f2 = self.auxiliaries[i][k]
+                    self.restrict(f1=f1, f2=f2)
+                    self.decomp[i].share_halos(f2)
+            if i not in self.smooth_args:
+                self.smooth_args[i] = {**self.unknowns[i], **self.rhos[i],
+                                       **self.auxiliaries[i],
+                                       **self.tmp[i]}
+                self.smooth_args[i]["dx"] = np.array(self.dx[i])
+            if i not in self.resid_args:
+                self.resid_args[i] = {**self.unknowns[i], **self.rhos[i],
+                                      **self.auxiliaries[i],
+                                      **self.resid[i]}
+                self.resid_args[i]["dx"] = np.array(self.dx[i])
+
+    def __call__(self, decomp0, queue=None, dx0=None, cycle=None, **kwargs):
+        """Execute a multigrid cycle (default: V(25,50) down to 8³;
+        reference multigrid/__init__.py:397-439)."""
+        if dx0 is None and queue is not None and \
+                not hasattr(queue, "rank_shape"):
+            dx0 = queue
+            queue = None
+        if cycle is None:
+            grid_shape = tuple(n * p for n, p in
+                               zip(decomp0.rank_shape, decomp0.proc_shape))
+            depth = int(np.log2(min(grid_shape) / 8))
+            cycle = v_cycle(25, 50, depth)
+
+        depth = max(i for i, _nu in cycle)
+        self.setup(decomp0, dx0=dx0, depth=depth, **kwargs)
+
+        nu0 = cycle[0][1]
+        level_errors = self.smooth(0, nu0)
+
+        previous = 0
+        for i, nu in cycle[1:]:
+            if i == previous + 1:
+                self.transfer_down(i=i)
+            elif i == previous - 1:
+                self.transfer_up(i=i)
+            else:
+                raise ValueError("consecutive levels must be spaced by one")
+            level_errors += self.smooth(i, nu)
+            previous = i
+        return level_errors
 
 
-class MultiGridSolver:
-    def __init__(self, *a, **kw):
-        raise NotImplementedError("linear multigrid: in progress")
+class MultiGridSolver(FullApproximationScheme):
+    """Linear multigrid: transfers the residual instead of the full
+    approximation (reference multigrid/__init__.py:442-487)."""
+
+    def transfer_down(self, queue=None, i=None):
+        if i is None:
+            i = queue
+        self.solver.residual(**self.resid_args[i - 1])
+        for f, rho in self.solver.f_to_rho_dict.items():
+            r1 = self.resid[i - 1]["r_" + f]
+            self.decomp[i - 1].share_halos(r1)
+            r2 = self.rhos[i][rho]
+            self.restrict(f1=r1, f2=r2)
+            self.decomp[i].share_halos(r2)
+        # coarse unknowns start at zero (correction equation)
+        for f2 in self.unknowns[i].values():
+            f2.zero_()
+
+    def transfer_up(self, queue=None, i=None):
+        if i is None:
+            i = queue
+        for key, f1 in self.unknowns[i].items():
+            f2 = self.unknowns[i + 1][key]
+            self.interpolate_and_correct(f1=f1, f2=f2)
+            self.decomp[i].share_halos(f1)

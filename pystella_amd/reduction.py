@@ -150,8 +150,10 @@ class Reduction:
             local = self._local_hip(env, rank_shape)
         else:
             local = self._local_torch(env, rank_shape)
+        return self._combine(local, rank_shape)
 
-        # one packed allreduce per op class
+    def _combine(self, local, rank_shape):
+        """One packed allreduce per op class, averaging, callback."""
         ops = [op for _, _, _, op in self.flat]
         results = list(local)
         if self.decomp.nranks > 1:

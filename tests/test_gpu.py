@@ -191,3 +191,43 @@ def test_native_extension_is_loaded():
     e = ext()
     assert e.device_count() >= 1
     assert "gfx95" in e.arch()
+
+
+@requires_gpu
+def test_fused_lap_reduction_vs_unfused(grid_shape=(32, 32, 32), h=2):
+    from pystella_amd.fusion import FusedLaplacianReduction
+    from pystella_amd.sectors import get_rho_and_p
+
+    def potential(f):
+        phi, chi = f[0], f[1]
+        return phi**2 / 2 + phi**2 * chi**2 / 4
+
+    sector = ps.ScalarSector(2, potential=potential)
+    decomp = ps.DomainDecomposition((1, 1, 1), h, rank_shape=grid_shape)
+    dx = (0.1, 0.1, 0.1)
+    derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=grid_shape)
+    pad = tuple(n + 2 * h for n in grid_shape)
+    torch.manual_seed(7)
+    f = torch.rand((2,) + pad, dtype=torch.float64).cuda()
+    dfdt = torch.rand((2,) + pad, dtype=torch.float64).cuda()
+    lap_u = torch.zeros((2,) + grid_shape, dtype=torch.float64,
+                        device="cuda")
+    lap_fz = torch.zeros_like(lap_u)
+    a = np.array([1.3])
+
+    gs = float(np.prod(grid_shape))
+    unfused = ps.Reduction(decomp, sector, halo_shape=h, grid_size=gs,
+                           callback=get_rho_and_p)
+    derivs(fx=f, lap=lap_u)
+    out_u = unfused(f=f, dfdt=dfdt, lap_f=lap_u, a=a)
+
+    fused = FusedLaplacianReduction(decomp, sector, derivs, halo_shape=h,
+                                    grid_size=gs, callback=get_rho_and_p)
+    out_f = fused(f=f, dfdt=dfdt, lap_f=lap_fz, a=a)
+    torch.cuda.synchronize()
+
+    assert (lap_fz - lap_u).abs().max().item() < 1e-13
+    for k in ("kinetic", "potential", "gradient"):
+        assert np.allclose(out_u[k], out_f[k], rtol=1e-12), k
+    assert abs(out_u["total"] - out_f["total"]) < 1e-12 * abs(
+        out_u["total"])

@@ -91,3 +91,38 @@ def test_scalar_sector_energy(grid_shape=(12, 12, 12)):
     assert abs(out["potential"][0] - pot) < 1e-12
     total = sum(kin) + sum(grad) + pot
     assert abs(out["total"] - total) < 1e-11
+
+
+def test_fused_lap_reduction_cpu(grid_shape=(16, 16, 16)):
+    """CPU path of the fused lap+energy component equals derivs +
+    Reduction composition."""
+    from pystella_amd.fusion import FusedLaplacianReduction
+    from pystella_amd.sectors import get_rho_and_p
+    h = 2
+    pad = tuple(n + 2 * h for n in grid_shape)
+    decomp = ps.DomainDecomposition((1, 1, 1), h, rank_shape=grid_shape)
+    dx = (0.2, 0.2, 0.2)
+    derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=grid_shape)
+
+    def potential(f):
+        return f[0]**2 / 2
+
+    sector = ps.ScalarSector(2, potential=potential)
+    torch.manual_seed(9)
+    f = torch.rand((2,) + pad, dtype=torch.float64)
+    dfdt = torch.rand((2,) + pad, dtype=torch.float64)
+    lap1 = torch.zeros((2,) + grid_shape, dtype=torch.float64)
+    lap2 = torch.zeros_like(lap1)
+    gs = float(np.prod(grid_shape))
+
+    derivs(fx=f, lap=lap1)
+    unfused = ps.Reduction(decomp, sector, halo_shape=h, grid_size=gs,
+                           callback=get_rho_and_p)
+    out_u = unfused(f=f, dfdt=dfdt, lap_f=lap1, a=np.array([1.1]))
+
+    fused = FusedLaplacianReduction(decomp, sector, derivs, halo_shape=h,
+                                    grid_size=gs, callback=get_rho_and_p)
+    out_f = fused(f=f, dfdt=dfdt, lap_f=lap2, a=np.array([1.1]))
+
+    assert (lap1 - lap2).abs().max().item() < 1e-14
+    assert abs(out_u["total"] - out_f["total"]) < 1e-13
