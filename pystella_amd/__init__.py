@@ -23,6 +23,7 @@ from pystella_amd.field import (  # noqa: F401
 )
 from pystella_amd.decomp import DomainDecomposition, init_distributed  # noqa: F401
 from pystella_amd.elementwise import ElementWiseMap  # noqa: F401
+from pystella_amd.stencil import Stencil, StreamingStencil  # noqa: F401
 from pystella_amd.reduction import Reduction, FieldStatistics  # noqa: F401
 from pystella_amd.histogram import Histogrammer, FieldHistogrammer  # noqa: F401
 from pystella_amd.derivs import (  # noqa: F401

@@ -25,6 +25,11 @@ _EXT = None
 def ext():
     global _EXT
     if _EXT is None:
+        from pystella_amd.backend import build as _build
+        if _build.needs_rebuild():
+            # sources newer than the shipped .so (e.g. edited after the
+            # last build) — rebuild in place rather than run stale code
+            _build.build_extension()
         try:
             from pystella_amd import _C
         except ImportError as e:
@@ -124,9 +129,9 @@ class JitElementwise:
 
 
 def get_elementwise_kernel(map_dict, tmp_instructions, field_args,
-                           scalar_names, halo, rank_shape):
+                           scalar_names, halo, rank_shape, name="ew_map"):
     return JitElementwise(map_dict, tmp_instructions, field_args,
-                          scalar_names, halo, rank_shape)
+                          scalar_names, halo, rank_shape, name=name)
 
 
 # ---------------------------------------------------------------------------

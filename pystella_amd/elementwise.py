@@ -41,7 +41,8 @@ class ElementWiseMap:
     """
 
     def __init__(self, map_dict, tmp_instructions=None, halo_shape=0,
-                 rank_shape=None, args=None, **kwargs):
+                 rank_shape=None, args=None, name="ew_map", **kwargs):
+        self.name = name
         self.map_dict = dict(map_dict)
         self.tmp_instructions = dict(tmp_instructions or {})
         self.halo_shape = ((halo_shape,) * 3
@@ -116,5 +117,6 @@ class ElementWiseMap:
                 self._hip_kernel.rank_shape != rank_shape:
             self._hip_kernel = get_elementwise_kernel(
                 self.map_dict, self.tmp_instructions, self.field_args,
-                sorted(self.scalar_names), self.halo_shape, rank_shape)
+                sorted(self.scalar_names), self.halo_shape, rank_shape,
+                name=self.name)
         self._hip_kernel(env)

@@ -124,7 +124,7 @@ class RungeKuttaStepper(Stepper):
                     rk_dict[lhs] = val
             steps.append(ElementWiseMap(
                 rk_dict, tmp_instructions=tmp, halo_shape=self.halo_shape,
-                rank_shape=self.rank_shape,
+                rank_shape=self.rank_shape, name=f"rk_stage{stage}",
                 fixed_parameters=fixed_parameters, **kwargs))
         return steps
 
@@ -256,7 +256,7 @@ class LowStorageRKStepper(Stepper):
                 rk_dict[key] = key + self._B[stage] * k_acc
             steps.append(ElementWiseMap(
                 rk_dict, tmp_instructions=tmp, halo_shape=self.halo_shape,
-                rank_shape=self.rank_shape,
+                rank_shape=self.rank_shape, name=f"rk_stage{stage}",
                 fixed_parameters=fixed_parameters, **kwargs))
         self.tmp_arrays = {}
         return steps

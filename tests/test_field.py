@@ -107,3 +107,26 @@ def test_stringify():
     f = Field("f", offset="h")
     assert "f" in str(f + 1)
     assert str(var("a") * var("b"))
+
+
+def test_stencil_map():
+    """Stencil kernels: shifted reads of padded fields
+    (analogue of reference test/test_stencil.py)."""
+    import pystella_amd as ps
+    from pystella_amd.field import shift_fields
+    h = 1
+    n = (8, 8, 8)
+    f = Field("f", offset="h")
+    out = Field("out", offset=0)
+    # 6-point neighbor sum
+    expr = sum(shift_fields(f, tuple(d * s for d in dirn))
+               for dirn in ((1, 0, 0), (0, 1, 0), (0, 0, 1))
+               for s in (1, -1))
+    knl = ps.Stencil({out: expr}, halo_shape=h, rank_shape=n)
+    t = torch.arange((8 + 2)**3, dtype=torch.float64).reshape(10, 10, 10)
+    o = torch.zeros(n, dtype=torch.float64)
+    knl(f=t, out=o)
+    expect = (t[2:, 1:-1, 1:-1] + t[:-2, 1:-1, 1:-1]
+              + t[1:-1, 2:, 1:-1] + t[1:-1, :-2, 1:-1]
+              + t[1:-1, 1:-1, 2:] + t[1:-1, 1:-1, :-2])
+    assert torch.allclose(o, expect)
