@@ -82,25 +82,41 @@ def _resolve_scalar(env, key):
 
 ELEMENTWISE_TEMPLATE = """{defines}
 {preamble}
-extern "C" __global__ __launch_bounds__(256) void {name}(
+extern "C" __global__ __launch_bounds__(TBZ * TBY) void {name}(
     {params})
 {{
-    const int k = blockIdx.x * 64 + (threadIdx.x % 64);
-    const int j = blockIdx.y * 4 + (threadIdx.x / 64);
+    const int k = blockIdx.x * TBZ + (threadIdx.x % TBZ);
+    const int j = blockIdx.y * TBY + (threadIdx.x / TBZ);
     if (k >= NZ || j >= NY) return;
-    for (int i = 0; i < NX; ++i) {{
+    const int i0 = blockIdx.z * XCHUNK;
+    const int i1 = (i0 + XCHUNK < NX) ? i0 + XCHUNK : NX;
+    for (int i = i0; i < i1; ++i) {{
         {body}
     }}
 }}
 """
 
 
+def _tile_defines(tile, rank_shape):
+    tbz, tby, xchunk = tile
+    return (f"#define TBZ {tbz}\n#define TBY {tby}\n"
+            f"#define XCHUNK {xchunk}\n")
+
+
+def _tile_grid(tile, rank_shape):
+    tbz, tby, xchunk = tile
+    nx, ny, nz = rank_shape
+    return ((nz + tbz - 1) // tbz, (ny + tby - 1) // tby,
+            (nx + xchunk - 1) // xchunk)
+
+
 class JitElementwise:
     """Compiled fused per-site map over the interior grid."""
 
     def __init__(self, map_dict, tmp_instructions, field_args, scalar_names,
-                 halo, rank_shape, name="ew_map"):
+                 halo, rank_shape, name="ew_map", tile=(64, 4, 64)):
         self.rank_shape = tuple(rank_shape)
+        self.tile = tile
         self.field_args = [fa for fa in field_args if fa.spatial]
         cg = Codegen(field_args, halo, rank_shape)
         body = cg.emit_statements(map_dict, tmp_instructions)
@@ -109,14 +125,14 @@ class JitElementwise:
         dbl_params = ", ".join(f"double {c}" for c, _ in cg.scalars)
         params = ", ".join(x for x in (ptr_params, dbl_params) if x)
         src = ELEMENTWISE_TEMPLATE.format(
-            defines=geometry_defines(halo, rank_shape), preamble=PREAMBLE,
-            name=name, params=params, body=body)
+            defines=geometry_defines(halo, rank_shape)
+            + _tile_defines(tile, rank_shape),
+            preamble=PREAMBLE, name=name, params=params, body=body)
         self.source = src
         self.scalar_keys = [k for _, k in cg.scalars]
         self.key = ext().jit_compile(src, name)
-
-        nx, ny, nz = rank_shape
-        self.grid = ((nz + 63) // 64, (ny + 3) // 4)
+        self.grid = _tile_grid(tile, rank_shape)
+        self.block = tile[0] * tile[1]
 
     def __call__(self, env):
         ptrs = []
@@ -124,42 +140,32 @@ class JitElementwise:
             t = _check_tensor(fa.name, env[fa.name])
             ptrs.append(t.data_ptr())
         doubles = [_resolve_scalar(env, k) for k in self.scalar_keys]
-        ext().jit_launch(self.key, self.grid[0], self.grid[1], 1,
-                         256, 1, 1, 0, _stream(), ptrs, [], doubles)
+        ext().jit_launch(self.key, self.grid[0], self.grid[1],
+                         self.grid[2], self.block, 1, 1, 0, _stream(),
+                         ptrs, [], doubles)
 
 
 def get_elementwise_kernel(map_dict, tmp_instructions, field_args,
-                           scalar_names, halo, rank_shape, name="ew_map"):
+                           scalar_names, halo, rank_shape, name="ew_map",
+                           tile=(64, 4, 64)):
     return JitElementwise(map_dict, tmp_instructions, field_args,
-                          scalar_names, halo, rank_shape, name=name)
+                          scalar_names, halo, rank_shape, name=name,
+                          tile=tile)
 
 
 # ---------------------------------------------------------------------------
 # JIT'd simultaneous reductions
 
-REDUCTION_TEMPLATE = """{defines}
-{preamble}
-#define NRED {nred}
-extern "C" __global__ __launch_bounds__(256) void {name}(
-    {params})
-{{
-    double acc[NRED];
-    {init}
-    const int k = blockIdx.x * 64 + (threadIdx.x % 64);
-    const int j = blockIdx.y * 4 + (threadIdx.x / 64);
-    if (k < NZ && j < NY) {{
-        for (int i = 0; i < NX; ++i) {{
-            {body}
-        }}
-    }}
-    __shared__ double sd[256];
-    const int nblk = gridDim.x * gridDim.y;
-    const int bid = blockIdx.y * gridDim.x + blockIdx.x;
+REDUCTION_TAIL = """
+    __shared__ double sd[TBZ * TBY];
+    const int nblk = gridDim.x * gridDim.y * gridDim.z;
+    const int bid = (blockIdx.z * gridDim.y + blockIdx.y) * gridDim.x
+                    + blockIdx.x;
     for (int r = 0; r < NRED; ++r) {{
         sd[threadIdx.x] = acc[r];
         __syncthreads();
-        for (int s = 128; s > 0; s >>= 1) {{
-            if (threadIdx.x < s)
+        for (int s = (TBZ * TBY) / 2; s > 0; s >>= 1) {{
+            if ((int)threadIdx.x < s)
                 sd[threadIdx.x] = COMBINE(r, sd[threadIdx.x],
                                           sd[threadIdx.x + s]);
             __syncthreads();
@@ -170,6 +176,25 @@ extern "C" __global__ __launch_bounds__(256) void {name}(
     }}
 }}
 """
+
+REDUCTION_TEMPLATE = """{defines}
+{preamble}
+#define NRED {nred}
+extern "C" __global__ __launch_bounds__(TBZ * TBY) void {name}(
+    {params})
+{{
+    double acc[NRED];
+    {init}
+    const int k = blockIdx.x * TBZ + (threadIdx.x % TBZ);
+    const int j = blockIdx.y * TBY + (threadIdx.x / TBZ);
+    const int i0 = blockIdx.z * XCHUNK;
+    const int i1 = (i0 + XCHUNK < NX) ? i0 + XCHUNK : NX;
+    if (k < NZ && j < NY) {{
+        for (int i = i0; i < i1; ++i) {{
+            {body}
+        }}
+    }}
+""" + REDUCTION_TAIL
 
 _OP_INIT = {"sum": "0.0", "avg": "0.0", "prod": "1.0",
             "max": "-1.0e308", "min": "1.0e308"}
@@ -182,8 +207,9 @@ class JitReduction:
     finished with torch ops + one packed allreduce by the caller."""
 
     def __init__(self, entries, field_args, scalar_names, halo, rank_shape,
-                 name="reduce_map"):
+                 name="reduce_map", tile=(64, 4, 64)):
         self.rank_shape = tuple(rank_shape)
+        self.tile = tile
         self.entries = entries
         self.field_args = [fa for fa in field_args if fa.spatial]
         nred = len(entries)
@@ -210,6 +236,7 @@ class JitReduction:
             ptr_params, "double* __restrict__ partials", dbl_params) if x)
 
         defines = geometry_defines(halo, rank_shape)
+        defines += _tile_defines(tile, rank_shape)
         defines += ("#define COMBINE(r, a, b) (" + combine + ")\n")
         src = REDUCTION_TEMPLATE.format(
             defines=defines, preamble=PREAMBLE, nred=nred, name=name,
@@ -219,10 +246,9 @@ class JitReduction:
         self.source = src
         self.scalar_keys = [k for _, k in cg.scalars]
         self.key = ext().jit_compile(src, name)
-
-        nx, ny, nz = rank_shape
-        self.grid = ((nz + 63) // 64, (ny + 3) // 4)
-        self.nblk = self.grid[0] * self.grid[1]
+        self.grid = _tile_grid(tile, rank_shape)
+        self.block = tile[0] * tile[1]
+        self.nblk = self.grid[0] * self.grid[1] * self.grid[2]
         self._partials = None
 
     def _finish(self, dev):
@@ -255,15 +281,16 @@ class JitReduction:
             self._partials = torch.empty((nred, self.nblk),
                                          dtype=torch.float64, device=dev)
         doubles = [_resolve_scalar(env, k) for k in self.scalar_keys]
-        ext().jit_launch(self.key, self.grid[0], self.grid[1], 1,
-                         256, 1, 1, 0, _stream(),
+        ext().jit_launch(self.key, self.grid[0], self.grid[1],
+                         self.grid[2], self.block, 1, 1, 0, _stream(),
                          ptrs + [self._partials.data_ptr()], [], doubles)
         return self._finish(dev)
 
 
 def get_reduction_kernel(entries, field_args, scalar_names, halo,
-                         rank_shape):
-    return JitReduction(entries, field_args, scalar_names, halo, rank_shape)
+                         rank_shape, tile=(64, 4, 64)):
+    return JitReduction(entries, field_args, scalar_names, halo,
+                        rank_shape, tile=tile)
 
 
 # ---------------------------------------------------------------------------
@@ -440,13 +467,15 @@ LAPRED_TEMPLATE = """{defines}
 {preamble}
 #define NRED {nred}
 #define NF {nf}
-extern "C" __global__ __launch_bounds__(256) void {name}(
+extern "C" __global__ __launch_bounds__(TBZ * TBY) void {name}(
     {params})
 {{
     double acc[NRED];
     {init}
-    const int k = blockIdx.x * 64 + (threadIdx.x % 64);
-    const int j = blockIdx.y * 4 + (threadIdx.x / 64);
+    const int k = blockIdx.x * TBZ + (threadIdx.x % TBZ);
+    const int j = blockIdx.y * TBY + (threadIdx.x / TBZ);
+    const int i0 = blockIdx.z * XCHUNK;
+    const int i1 = (i0 + XCHUNK < NX) ? i0 + XCHUNK : NX;
     if (k < NZ && j < NY) {{
         const long sx = PSY * PSZ;
         double ring[NF][2 * H + 1];
@@ -456,9 +485,9 @@ extern "C" __global__ __launch_bounds__(256) void {name}(
                                + (long)(j + H) * PSZ + (k + H);
             #pragma unroll
             for (int p = 0; p < 2 * H; ++p)
-                ring[fld][p] = fp[(long)p * sx];
+                ring[fld][p] = fp[(long)(i0 + p) * sx];
         }}
-        for (int i = 0; i < NX; ++i) {{
+        for (int i = i0; i < i1; ++i) {{
             double lapv[NF];
             #pragma unroll
             for (int fld = 0; fld < NF; ++fld) {{
@@ -480,24 +509,7 @@ extern "C" __global__ __launch_bounds__(256) void {name}(
                     ring[fld][p] = ring[fld][p + 1];
         }}
     }}
-    __shared__ double sd[256];
-    const int nblk = gridDim.x * gridDim.y;
-    const int bid = blockIdx.y * gridDim.x + blockIdx.x;
-    for (int r = 0; r < NRED; ++r) {{
-        sd[threadIdx.x] = acc[r];
-        __syncthreads();
-        for (int s = 128; s > 0; s >>= 1) {{
-            if (threadIdx.x < s)
-                sd[threadIdx.x] = COMBINE(r, sd[threadIdx.x],
-                                          sd[threadIdx.x + s]);
-            __syncthreads();
-        }}
-        if (threadIdx.x == 0)
-            partials[(long)r * nblk + bid] = sd[0];
-        __syncthreads();
-    }}
-}}
-"""
+""" + REDUCTION_TAIL
 
 
 class _LapCodegen(Codegen):
@@ -523,9 +535,11 @@ class JitLapReduction:
     """Fused lap-stencil + multi-quantity reduction (see module note)."""
 
     def __init__(self, entries, field_args, scalar_names, halo, rank_shape,
-                 dx, nf, f_name="f", lap_name="lap_f", name="lapred_map"):
+                 dx, nf, f_name="f", lap_name="lap_f", name="lapred_map",
+                 tile=(64, 4, 64)):
         from pystella_amd.derivs import _LAP_COEFS
         self.rank_shape = tuple(rank_shape)
+        self.tile = tile
         self.entries = entries
         h = max(halo) if isinstance(halo, (tuple, list)) else halo
         self.nf = nf
@@ -567,6 +581,7 @@ class JitLapReduction:
             ptr_params, "double* __restrict__ partials", dbl_params) if x)
 
         defines = geometry_defines(halo, rank_shape)
+        defines += _tile_defines(tile, rank_shape)
         defines += f"#define COMBINE(r, a, b) ({combine})\n"
         defines += f"#define LAPC0 ({lapc0!r})\n"
         src = LAPRED_TEMPLATE.format(
@@ -578,10 +593,9 @@ class JitLapReduction:
         self.source = src
         self.scalar_keys = [k for _, k in cg.scalars]
         self.key = ext().jit_compile(src, name)
-
-        nx, ny, nz = rank_shape
-        self.grid = ((nz + 63) // 64, (ny + 3) // 4)
-        self.nblk = self.grid[0] * self.grid[1]
+        self.grid = _tile_grid(tile, rank_shape)
+        self.block = tile[0] * tile[1]
+        self.nblk = self.grid[0] * self.grid[1] * self.grid[2]
         self._partials = None
 
     _finish = JitReduction._finish
@@ -600,14 +614,14 @@ class JitLapReduction:
             self._partials = torch.empty((nred, self.nblk),
                                          dtype=torch.float64, device=dev)
         doubles = [_resolve_scalar(env, k) for k in self.scalar_keys]
-        ext().jit_launch(self.key, self.grid[0], self.grid[1], 1,
-                         256, 1, 1, 0, _stream(),
+        ext().jit_launch(self.key, self.grid[0], self.grid[1],
+                         self.grid[2], self.block, 1, 1, 0, _stream(),
                          ptrs + [self._partials.data_ptr()], [], doubles)
         return self._finish(dev)
 
 
 def get_lap_reduction_kernel(entries, field_args, scalar_names, halo,
                              rank_shape, dx, nf, f_name="f",
-                             lap_name="lap_f"):
+                             lap_name="lap_f", tile=(64, 4, 64)):
     return JitLapReduction(entries, field_args, scalar_names, halo,
-                           rank_shape, dx, nf, f_name, lap_name)
+                           rank_shape, dx, nf, f_name, lap_name, tile=tile)

@@ -20,6 +20,7 @@
 
 #include <hip/hip_runtime.h>
 #include <cstdint>
+#include <cstdlib>
 
 namespace {
 
@@ -82,12 +83,15 @@ __global__ __launch_bounds__(BZ * BY) void gradlap_knl(
     const double *__restrict__ f, double *__restrict__ lap,
     double *__restrict__ pdx, double *__restrict__ pdy,
     double *__restrict__ pdz, int64_t g_fstride, int nx, int ny, int nz,
+    int nxch, int xchunk,
     double inv_dx, double inv_dy, double inv_dz,
     double inv_dx2, double inv_dy2, double inv_dz2)
 {
     const int k = blockIdx.x * BZ + (threadIdx.x % BZ);
     const int j = blockIdx.y * BY + (threadIdx.x / BZ);
-    const int fld = blockIdx.z;
+    const int fld = blockIdx.z / nxch;
+    const int i0 = (blockIdx.z % nxch) * xchunk;
+    const int i1 = (i0 + xchunk < nx) ? i0 + xchunk : nx;
     if (k >= nz || j >= ny) return;
 
     const int64_t psz = nz + 2 * H;
@@ -112,9 +116,9 @@ __global__ __launch_bounds__(BZ * BY) void gradlap_knl(
     // register ring r[p] holds f at x = i - H + p (center plane p = H)
     double r[2 * H + 1];
 #pragma unroll
-    for (int p = 0; p < 2 * H; ++p) r[p] = fp[(int64_t)p * sx];
+    for (int p = 0; p < 2 * H; ++p) r[p] = fp[(int64_t)(i0 + p) * sx];
 
-    for (int i = 0; i < nx; ++i) {
+    for (int i = i0; i < i1; ++i) {
         r[2 * H] = fp[(int64_t)(i + 2 * H) * sx];
         const double c = r[H];
         const double *cp = fp + (int64_t)(i + H) * sx;  // center plane
@@ -155,11 +159,13 @@ __global__ __launch_bounds__(BZ * BY) void gradlap_knl(
 template <int H, int AXIS, bool ACCUM>
 __global__ __launch_bounds__(BZ * BY) void pd_knl(
     const double *__restrict__ f, double *__restrict__ out,
-    int nx, int ny, int nz, double inv_d)
+    int nx, int ny, int nz, int nxch, int xchunk, double inv_d)
 {
     const int k = blockIdx.x * BZ + (threadIdx.x % BZ);
     const int j = blockIdx.y * BY + (threadIdx.x / BZ);
-    const int fld = blockIdx.z;
+    const int fld = blockIdx.z / nxch;
+    const int i0 = (blockIdx.z % nxch) * xchunk;
+    const int i1 = (i0 + xchunk < nx) ? i0 + xchunk : nx;
     if (k >= nz || j >= ny) return;
 
     const int64_t psz = nz + 2 * H;
@@ -174,7 +180,7 @@ __global__ __launch_bounds__(BZ * BY) void pd_knl(
     double *op = out + (int64_t)fld * uvol + (int64_t)j * nz + k;
     const int64_t so = (int64_t)ny * nz;
 
-    for (int i = 0; i < nx; ++i) {
+    for (int i = i0; i < i1; ++i) {
         const double *cp = fp + (int64_t)(i + H) * sx;
         double g = 0.;
 #pragma unroll
@@ -189,9 +195,19 @@ __global__ __launch_bounds__(BZ * BY) void pd_knl(
     }
 }
 
-inline dim3 tile_grid(int ny, int nz, int nf)
+// XCHUNK: split the x-march into chunks for more resident blocks
+// (latency hiding); tunable via PYSTELLA_XCHUNK host-side.
+inline int xchunk_size(int nx)
 {
-    return dim3((nz + BZ - 1) / BZ, (ny + BY - 1) / BY, nf);
+    const char *env = getenv("PYSTELLA_XCHUNK");
+    int c = env ? atoi(env) : 64;
+    if (c <= 0 || c > nx) c = nx;
+    return c;
+}
+
+inline dim3 tile_grid(int ny, int nz, int nf, int nxch)
+{
+    return dim3((nz + BZ - 1) / BZ, (ny + BY - 1) / BY, nf * nxch);
 }
 
 }  // namespace
@@ -211,7 +227,9 @@ extern "C" int pystella_gradlap(
     double dx, double dy, double dz, void *stream_)
 {
     hipStream_t stream = (hipStream_t)stream_;
-    const dim3 grid = tile_grid(ny, nz, nf);
+    const int xchunk = xchunk_size(nx);
+    const int nxch = (nx + xchunk - 1) / xchunk;
+    const dim3 grid = tile_grid(ny, nz, nf, nxch);
     const dim3 block(BZ * BY);
     const double ix = 1. / dx, iy = 1. / dy, iz = 1. / dz;
     const double ix2 = ix * ix, iy2 = iy * iy, iz2 = iz * iz;
@@ -222,18 +240,18 @@ extern "C" int pystella_gradlap(
         if (do_lap && do_grad)
             hipLaunchKernelGGL((gradlap_knl<H, true, true>), grid, block, 0,
                                stream, f, lap, pdx, pdy, pdz,
-                               (int64_t)g_fstride, nx, ny, nz,
-                               ix, iy, iz, ix2, iy2, iz2);
+                               (int64_t)g_fstride, nx, ny, nz, nxch,
+                               xchunk, ix, iy, iz, ix2, iy2, iz2);
         else if (do_lap)
             hipLaunchKernelGGL((gradlap_knl<H, true, false>), grid, block, 0,
                                stream, f, lap, pdx, pdy, pdz,
-                               (int64_t)g_fstride, nx, ny, nz,
-                               ix, iy, iz, ix2, iy2, iz2);
+                               (int64_t)g_fstride, nx, ny, nz, nxch,
+                               xchunk, ix, iy, iz, ix2, iy2, iz2);
         else
             hipLaunchKernelGGL((gradlap_knl<H, false, true>), grid, block, 0,
                                stream, f, lap, pdx, pdy, pdz,
-                               (int64_t)g_fstride, nx, ny, nz,
-                               ix, iy, iz, ix2, iy2, iz2);
+                               (int64_t)g_fstride, nx, ny, nz, nxch,
+                               xchunk, ix, iy, iz, ix2, iy2, iz2);
     });
     return (int)hipGetLastError();
 }
@@ -243,24 +261,26 @@ extern "C" int pystella_pd(
     int nx, int ny, int nz, int nf, double d, void *stream_)
 {
     hipStream_t stream = (hipStream_t)stream_;
-    const dim3 grid = tile_grid(ny, nz, nf);
+    const int xchunk = xchunk_size(nx);
+    const int nxch = (nx + xchunk - 1) / xchunk;
+    const dim3 grid = tile_grid(ny, nz, nf, nxch);
     const dim3 block(BZ * BY);
     const double inv = 1. / d;
 
     DISPATCH_H(h, {
         switch (axis * 2 + (accum ? 1 : 0)) {
         case 0: hipLaunchKernelGGL((pd_knl<H, 0, false>), grid, block, 0,
-                                   stream, f, out, nx, ny, nz, inv); break;
+                                   stream, f, out, nx, ny, nz, nxch, xchunk, inv); break;
         case 1: hipLaunchKernelGGL((pd_knl<H, 0, true>), grid, block, 0,
-                                   stream, f, out, nx, ny, nz, inv); break;
+                                   stream, f, out, nx, ny, nz, nxch, xchunk, inv); break;
         case 2: hipLaunchKernelGGL((pd_knl<H, 1, false>), grid, block, 0,
-                                   stream, f, out, nx, ny, nz, inv); break;
+                                   stream, f, out, nx, ny, nz, nxch, xchunk, inv); break;
         case 3: hipLaunchKernelGGL((pd_knl<H, 1, true>), grid, block, 0,
-                                   stream, f, out, nx, ny, nz, inv); break;
+                                   stream, f, out, nx, ny, nz, nxch, xchunk, inv); break;
         case 4: hipLaunchKernelGGL((pd_knl<H, 2, false>), grid, block, 0,
-                                   stream, f, out, nx, ny, nz, inv); break;
+                                   stream, f, out, nx, ny, nz, nxch, xchunk, inv); break;
         case 5: hipLaunchKernelGGL((pd_knl<H, 2, true>), grid, block, 0,
-                                   stream, f, out, nx, ny, nz, inv); break;
+                                   stream, f, out, nx, ny, nz, nxch, xchunk, inv); break;
         }
     });
     return (int)hipGetLastError();

@@ -80,7 +80,15 @@ class RayleighGenerator:
 
     def _post_process(self, fk):
         if self.fft.is_real:
-            make_hermitian(fk)
+            if tuple(fk.shape[:2]) == tuple(self.fft.grid_shape[:2]):
+                # single-rank layout: enforce exact Hermitian symmetry
+                # on the kz ∈ {0, Nyquist} planes
+                make_hermitian(fk)
+            # distributed pencils: the kz=0/Nyquist planes are spread
+            # over ranks; the c2r transform's z-axis symmetry plus
+            # corner-mode cleanup keeps the field real (matching the
+            # reference's pDFT path, which also skips make_hermitian:
+            # rayleigh.py:160-171)
             self.fft.zero_corner_modes(fk, only_imag=True)
         return fk
 

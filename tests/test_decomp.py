@@ -116,3 +116,33 @@ def test_uneven_split():
     assert (n, s) == (3, 4)
     n, s = ps.decomp.get_size_start(10, 3, 2)
     assert (n, s) == (3, 7)
+
+
+def _checkpoint_worker(rank, world_size, mode):
+    import torch
+    from pystella_amd.checkpoint import save_checkpoint, load_checkpoint
+    import tempfile
+    import os
+    h = 1
+    grid = (8, 8, 8)
+    decomp = ps.DomainDecomposition((world_size, 1, 1), h, grid_shape=grid)
+    rank_shape, _ = decomp.get_rank_shape_start(grid)
+    pad = tuple(n + 2 * h for n in rank_shape)
+    torch.manual_seed(rank)
+    f = torch.rand((2,) + pad, dtype=torch.float64)
+    decomp.share_halos(f)
+    path = os.path.join(tempfile.gettempdir(),
+                        f"ckpt_test_{mode}_{world_size}.pt")
+    save_checkpoint(path, decomp, {"f": f}, attrs={"t": 1.5}, mode=mode)
+    f2 = torch.zeros_like(f)
+    attrs = load_checkpoint(path, decomp, {"f": f2})
+    assert attrs["t"] == 1.5
+    assert torch.allclose(f, f2), f"rank {rank} mismatch"
+
+
+def test_checkpoint_gather():
+    run_distributed(_checkpoint_worker, 2, args=("gather",))
+
+
+def test_checkpoint_shard():
+    run_distributed(_checkpoint_worker, 2, args=("shard",))

@@ -45,3 +45,34 @@ def test_scalar_preheating_output(tmp_path):
     energy = store.read("energy", "total")
     assert energy.shape[0] >= 1
     assert np.isfinite(energy).all()
+
+
+def test_scalar_preheating_gws(tmp_path):
+    """Gravitational-wave sector end to end (CPU, small grid)."""
+    import scalar_preheating
+    os.chdir(tmp_path)
+    expand, energy = scalar_preheating.main(
+        ["--grid-shape", "12", "12", "12", "--end-time", "0.3",
+         "--device", "cpu", "--no-output", "--gravitational-waves"])
+    assert np.isfinite(expand.constraint(energy["total"]))
+
+
+def _preheating_dist_worker(rank, world_size):
+    import scalar_preheating
+    expand, energy = scalar_preheating.main(
+        ["--grid-shape", "16", "16", "16", "--proc-shape", "2", "1", "1",
+         "--end-time", "0.2", "--device", "cpu", "--no-output"])
+    # all ranks agree on the (allreduced) energy and expansion state
+    assert np.isfinite(energy["total"])
+    import pystella_amd as ps
+    import torch.distributed as dist
+    import torch
+    t = torch.tensor([energy["total"], float(expand.a[0])])
+    t0 = t.clone()
+    dist.broadcast(t0, src=0)
+    assert torch.allclose(t, t0), (rank, t, t0)
+
+
+def test_scalar_preheating_distributed():
+    from tests.conftest import run_distributed
+    run_distributed(_preheating_dist_worker, 2)

@@ -1,0 +1,119 @@
+"""Tile-shape sweep for the hot kernels on a live MI355X.
+
+Times the RK stage kernel, the fused lap+energy kernel and the AOT
+gradlap kernel across (TBZ, TBY, XCHUNK) tiles and prints a ranked
+table.  Winning defaults get baked into backend/hip.py / the
+PYSTELLA_XCHUNK env default.
+"""
+
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+sys.path.insert(0, ".")
+import pystella_amd as ps  # noqa: E402
+from pystella_amd.backend import hip as H  # noqa: E402
+from pystella_amd.sectors import get_rho_and_p  # noqa: E402
+from pystella_amd.reduction import Reduction  # noqa: E402
+
+
+def timeit(fn, n=10, warmup=2):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(n):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / n * 1e3
+
+
+def main(n=512):
+    device = torch.device("cuda", 0)
+    torch.cuda.set_device(device)
+    grid = (n, n, n)
+    h = 2
+    dx = tuple(5 / g for g in grid)
+    dt = 0.1 * min(dx)
+    nscalars = 2
+    sites = float(np.prod(grid))
+    GBms = sites * nscalars * 8 / 1e9  # GB per (field pass) per ms scale
+
+    decomp = ps.DomainDecomposition((1, 1, 1), h, grid_shape=grid)
+    pad = tuple(g + 2 * h for g in grid)
+
+    def potential(f):
+        return (1.2e-6**2 / 2 * f[0]**2
+                + 2.5e-7 / 2 * f[0]**2 * f[1]**2) / 1.2e-6**2
+
+    sector = ps.ScalarSector(nscalars, potential=potential)
+    gen = torch.Generator(device="cpu").manual_seed(1)
+    f = (0.19 + 1e-3 * torch.rand((nscalars,) + pad, dtype=torch.float64,
+                                  generator=gen)).to(device)
+    dfdt = torch.zeros_like(f)
+    lap = torch.zeros((nscalars,) + grid, dtype=torch.float64,
+                      device=device)
+    a = np.ones(1)
+    hub = np.zeros(1)
+
+    tiles = [(64, 4, 512), (64, 4, 128), (64, 4, 64), (64, 4, 32),
+             (64, 8, 64), (64, 8, 32), (128, 2, 64), (64, 2, 64),
+             (256, 1, 64), (64, 16, 32)]
+
+    # --- RK stage kernel sweep
+    stepper = ps.LowStorageRK54([sector], halo_shape=h, rank_shape=grid,
+                                dt=dt)
+    step1 = stepper.steps[1]      # stage 1 reads+writes everything
+    env = dict(a=a, hubble=hub, f=f, dfdt=dfdt, lap_f=lap,
+               f_tmp=torch.zeros((nscalars,) + grid, dtype=torch.float64,
+                                 device=device),
+               dfdt_tmp=torch.zeros((nscalars,) + grid,
+                                    dtype=torch.float64, device=device),
+               dt=dt)
+    print(f"== rk stage kernel (10 passes ~ {10*GBms:.1f} GB)")
+    results = []
+    for tile in tiles:
+        k = H.JitElementwise(step1.map_dict, step1.tmp_instructions,
+                             step1.field_args, [], (h,) * 3, grid,
+                             name=f"tune_st_{tile[0]}_{tile[1]}_{tile[2]}",
+                             tile=tile)
+        ms = timeit(lambda: k(env))
+        bw = 10 * GBms / ms
+        results.append((ms, tile))
+        print(f"  tile={tile}:  {ms:7.3f} ms   {bw:5.2f} TB/s")
+    results.sort()
+    print("  BEST:", results[0])
+
+    # --- fused lap+energy sweep
+    red = Reduction(decomp, sector, halo_shape=h, grid_size=sites,
+                    callback=get_rho_and_p, rank_shape=grid)
+    entries = [(e, o) for _, _, e, o in red.flat]
+    print(f"== fused lap+energy (6 passes ~ {6*GBms:.1f} GB)")
+    results = []
+    env2 = dict(f=f, dfdt=dfdt, lap_f=lap, a=a)
+    for tile in tiles:
+        k = H.JitLapReduction(
+            entries, red.field_args, [], (h,) * 3, grid, dx, nscalars,
+            name=f"tune_lr_{tile[0]}_{tile[1]}_{tile[2]}", tile=tile)
+        ms = timeit(lambda: k(env2))
+        bw = 6 * GBms / ms
+        results.append((ms, tile))
+        print(f"  tile={tile}:  {ms:7.3f} ms   {bw:5.2f} TB/s")
+    results.sort()
+    print("  BEST:", results[0])
+
+    # --- AOT gradlap XCHUNK sweep (env var)
+    derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=grid)
+    print(f"== AOT gradlap lap-only (4 passes ~ {4*GBms:.1f} GB)")
+    for xc in (512, 128, 64, 32):
+        os.environ["PYSTELLA_XCHUNK"] = str(xc)
+        ms = timeit(lambda: H.derivs(f, lap=lap, halo=(h,) * 3, dx=dx,
+                                     h=h))
+        print(f"  XCHUNK={xc}:  {ms:7.3f} ms   {4*GBms/ms:5.2f} TB/s")
+
+
+if __name__ == "__main__":
+    main(int(sys.argv[1]) if len(sys.argv) > 1 else 512)
