@@ -130,3 +130,12 @@ def test_stencil_map():
               + t[1:-1, 2:, 1:-1] + t[1:-1, :-2, 1:-1]
               + t[1:-1, 1:-1, 2:] + t[1:-1, 1:-1, :-2])
     assert torch.allclose(o, expect)
+
+
+def test_profiler_cpu():
+    from pystella_amd.profiling import Profiler
+    prof = Profiler(enabled=True, use_events=False)
+    with prof.region("work", bytes=8):
+        sum(range(1000))
+    rep = prof.report()
+    assert "work" in rep
