@@ -40,6 +40,9 @@ def main():
     parser.add_argument("--grid", type=int, default=512)
     parser.add_argument("--halo", type=int, default=2)
     parser.add_argument("--device", default=None)
+    parser.add_argument("--no-fuse", action="store_true",
+                        help="reference-structure loop (separate lap "
+                             "array + unfused stage kernels)")
     p = parser.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -74,13 +77,23 @@ def main():
         return (mphi**2 / 2 * phi**2 + gsq / 2 * phi**2 * chi**2) / mphi**2
 
     sector = ps.ScalarSector(nscalars, potential=potential)
-    stepper = ps.LowStorageRK54([sector], halo_shape=h,
-                                rank_shape=rank_shape, dt=dt)
     derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=rank_shape)
-    from pystella_amd.fusion import FusedLaplacianReduction
+    from pystella_amd.fusion import (
+        FusedLaplacianReduction, StencilRKStepper)
+    if p.no_fuse:
+        stepper = ps.LowStorageRK54([sector], halo_shape=h,
+                                    rank_shape=rank_shape, dt=dt)
+    else:
+        # MI355X-optimized structure: stage kernels evaluate the
+        # Laplacian inline (ping-pong f); the energy reduction never
+        # materializes lap in HBM
+        stepper = StencilRKStepper(ps.LowStorageRK54, [sector], derivs,
+                                   halo_shape=h, rank_shape=rank_shape,
+                                   dt=dt)
     reduce_energy = FusedLaplacianReduction(
         decomp, sector, derivs, halo_shape=h, callback=get_rho_and_p,
-        rank_shape=rank_shape, grid_size=grid_size)
+        rank_shape=rank_shape, grid_size=grid_size,
+        store_lap=p.no_fuse)
 
     gen = torch.Generator(device="cpu").manual_seed(7 + decomp.rank)
     f = (0.193 + 1e-3 * torch.rand((nscalars,) + pad, dtype=torch.float64,
@@ -88,14 +101,21 @@ def main():
     dfdt = (-0.142 + 1e-3 * torch.rand((nscalars,) + pad,
                                        dtype=torch.float64,
                                        generator=gen)).to(device)
-    lap_f = torch.zeros((nscalars,) + tuple(rank_shape),
-                        dtype=torch.float64, device=device)
+    arrays = {"f": f, "dfdt": dfdt}
+    if p.no_fuse:
+        arrays["lap_f"] = torch.zeros(
+            (nscalars,) + tuple(rank_shape), dtype=torch.float64,
+            device=device)
+    else:
+        arrays["f_next"] = torch.zeros_like(f)
 
     energy = None
 
     def compute_energy(a):
-        # fused: halo exchange + Laplacian stencil + energy reduction
-        return reduce_energy(f=f, dfdt=dfdt, lap_f=lap_f, a=np.array(a))
+        # fused: halo exchange + inline Laplacian stencil + energy
+        # reduction (one kernel)
+        kw = {k: v for k, v in arrays.items() if k != "f_next"}
+        return reduce_energy(a=np.array(a), **kw)
 
     energy = compute_energy(1.)
     expand = ps.Expansion(energy["total"], ps.LowStorageRK54, mpl=mpl)
@@ -103,8 +123,9 @@ def main():
     def step():
         nonlocal energy
         for s in range(stepper.num_stages):
-            stepper(s, a=expand.a, hubble=expand.hubble,
-                    f=f, dfdt=dfdt, lap_f=lap_f)
+            stepper(s, a=expand.a, hubble=expand.hubble, **arrays)
+            if not p.no_fuse:
+                arrays["f"], arrays["f_next"] =                     arrays["f_next"], arrays["f"]
             expand.step(s, energy["total"], energy["pressure"], dt)
             energy = compute_energy(expand.a)
 

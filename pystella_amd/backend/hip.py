@@ -498,8 +498,10 @@ extern "C" __global__ __launch_bounds__(TBZ * TBY) void {name}(
                 double la = ring[fld][H] * LAPC0;
                 {lap_terms}
                 lapv[fld] = la;
+#if STORE_LAP
                 {lapname}[(long)fld * UVOL + (((long)i * NY + j) * NZ + k)]
                     = la;
+#endif
             }}
             {body}
             #pragma unroll
@@ -536,10 +538,11 @@ class JitLapReduction:
 
     def __init__(self, entries, field_args, scalar_names, halo, rank_shape,
                  dx, nf, f_name="f", lap_name="lap_f", name="lapred_map",
-                 tile=(64, 4, 64)):
+                 tile=(64, 4, 64), store_lap=True):
         from pystella_amd.derivs import _LAP_COEFS
         self.rank_shape = tuple(rank_shape)
         self.tile = tile
+        self.store_lap = store_lap
         self.entries = entries
         h = max(halo) if isinstance(halo, (tuple, list)) else halo
         self.nf = nf
@@ -568,8 +571,10 @@ class JitLapReduction:
                 f" + (cp[{s}] + cp[-{s}])*{inv2[2]!r});")
         lapc0 = coefs[0] * (inv2[0] + inv2[1] + inv2[2])
 
-        # pointer params: stencil field first, then lap, then the rest
-        self.ptr_names = [f_name, lap_name] + sorted(
+        # pointer params: stencil field first, then lap (if stored),
+        # then the rest
+        self.ptr_names = ([f_name, lap_name] if store_lap
+                          else [f_name]) + sorted(
             fa.name for fa in field_args
             if fa.spatial and fa.name not in (f_name, lap_name))
         by_name = {fa.name: fa for fa in field_args}
@@ -584,6 +589,7 @@ class JitLapReduction:
         defines += _tile_defines(tile, rank_shape)
         defines += f"#define COMBINE(r, a, b) ({combine})\n"
         defines += f"#define LAPC0 ({lapc0!r})\n"
+        defines += f"#define STORE_LAP {1 if store_lap else 0}\n"
         src = LAPRED_TEMPLATE.format(
             defines=defines, preamble=PREAMBLE, nred=len(entries), nf=nf,
             name=name, params=params, fname=f_name, lapname=lap_name,
@@ -622,6 +628,8 @@ class JitLapReduction:
 
 def get_lap_reduction_kernel(entries, field_args, scalar_names, halo,
                              rank_shape, dx, nf, f_name="f",
-                             lap_name="lap_f", tile=(64, 4, 64)):
+                             lap_name="lap_f", tile=(64, 4, 64),
+                             store_lap=True):
     return JitLapReduction(entries, field_args, scalar_names, halo,
-                           rank_shape, dx, nf, f_name, lap_name, tile=tile)
+                           rank_shape, dx, nf, f_name, lap_name, tile=tile,
+                           store_lap=store_lap)

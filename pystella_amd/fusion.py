@@ -39,20 +39,29 @@ class FusedLaplacianReduction(Reduction):
     """
 
     def __init__(self, decomp, input, derivs, f_name="f", lap_name="lap_f",
-                 **kwargs):
+                 store_lap=True, share_halos=True, **kwargs):
         super().__init__(decomp, input, **kwargs)
         self.derivs = derivs
         self.f_name = f_name
         self.lap_name = lap_name
+        self.store_lap = store_lap
+        self.share = share_halos
         self._fused_kernel = None
 
     def __call__(self, queue=None, filter_args=False, **kwargs):
         f = kwargs[self.f_name]
-        lap = kwargs[self.lap_name]
-        self.decomp.share_halos(f)
+        if self.share:
+            self.decomp.share_halos(f)
         if not (isinstance(f, torch.Tensor) and f.is_cuda):
-            # CPU oracle: unfused compose
-            self.derivs.decomp = self.decomp
+            # CPU oracle: unfused compose (lap into a scratch array if
+            # the caller does not keep one)
+            lap = kwargs.get(self.lap_name)
+            if lap is None:
+                shape = f.shape[:-3] + tuple(
+                    n - 2 * hh for n, hh in zip(f.shape[-3:],
+                                                self.halo_shape))
+                lap = torch.zeros(shape, dtype=f.dtype, device=f.device)
+                kwargs[self.lap_name] = lap
             from itertools import product
             for s in product(*[range(n) for n in f.shape[:-3]]):
                 self.derivs._apply_lap_cpu(f[s], lap[s])
@@ -67,6 +76,138 @@ class FusedLaplacianReduction(Reduction):
                 [(expr, op) for _, _, expr, op in self.flat],
                 self.field_args, sorted(self.scalar_names),
                 self.halo_shape, rank_shape, self.derivs.dx, nf,
-                self.f_name, self.lap_name)
+                self.f_name, self.lap_name, store_lap=self.store_lap)
         local = self._fused_kernel(kwargs)
         return self._combine(local, rank_shape)
+
+
+class StencilRKStepper:
+    """Low-storage RK stepper whose stage kernels evaluate the Laplacian
+    *inline* from the finite-difference stencil instead of reading a
+    precomputed ``lap_f`` array (MI355X traffic optimization: the lap
+    array never exists in HBM — per stage this removes one full
+    write+read pass per unknown vs the reference structure,
+    examples/scalar_preheating.py:258-271).
+
+    Fields whose ``.lap`` appears in the equations of motion are
+    double-buffered (the stencil reads ``f``; the update writes
+    ``f_next``) to avoid intra-kernel races; call :meth:`swap` names
+    after each stage.  All other unknowns update in place.
+
+    Usage::
+
+        stepper = StencilRKStepper(ps.LowStorageRK54, [sector], derivs,
+                                   halo_shape=h, rank_shape=shape, dt=dt)
+        arrays = {"f": f, "f_next": f_next, "dfdt": dfdt}
+        for s in range(stepper.num_stages):
+            stepper(s, a=a, hubble=hub, **arrays)
+            arrays["f"], arrays["f_next"] = arrays["f_next"], arrays["f"]
+            decomp.share_halos(arrays["f"])
+            # ... energy reduction on arrays["f"] ...
+    """
+
+    def __init__(self, Stepper, input, derivs, halo_shape=0,
+                 rank_shape=None, dt=None, **kwargs):
+        from pystella_amd.field import (
+            DynamicField, Field, Subscript, substitute, collect_fields)
+        from pystella_amd.derivs import _LAP_COEFS, centered_diff
+        from pystella_amd.step import LowStorageRKStepper, _field_of
+        from pystella_amd.sectors import Sector
+
+        if not issubclass(Stepper, LowStorageRKStepper):
+            raise TypeError("StencilRKStepper requires a low-storage "
+                            "Stepper")
+        if isinstance(input, Sector):
+            rhs_dict = dict(input.rhs_dict)
+        elif isinstance(input, list):
+            rhs_dict = {}
+            for s in input:
+                rhs_dict.update(s.rhs_dict)
+        else:
+            rhs_dict = dict(input)
+
+        h = max(derivs.halo_shape)
+        dx = derivs.dx
+        coefs = _LAP_COEFS[h]
+
+        # find DynamicFields whose .lap is referenced; build inline
+        # stencil substitutions and the ping-pong name map
+        fields = collect_fields(list(rhs_dict.values()))
+        lap_names = {f.name for f in fields if f.name.startswith("lap_")}
+        self.pingpong = []
+        subs = {}
+        for key in rhs_dict:
+            f, outer = _field_of(key)
+            lap_name = f"lap_{f.name}"
+            if lap_name in lap_names:
+                self.pingpong.append(f.name)
+                lap_f = Field(lap_name, offset=0, shape=f.shape,
+                              indices=f.indices)
+                for fld in range(f.shape[0] if f.shape else 1):
+                    access = f[fld] if f.shape else f
+                    lap_expr = sum(
+                        centered_diff(access, coefs, direction=mu + 1,
+                                      order=2) * (1.0 / dx[mu] ** 2)
+                        for mu in range(3))
+                    lap_acc = lap_f[fld] if f.shape else lap_f
+                    subs[lap_acc] = lap_expr
+        self.pingpong = sorted(set(self.pingpong))
+        new_rhs = {k: substitute(v, subs) for k, v in rhs_dict.items()}
+
+        # redirect writes of ping-ponged fields to NAME_next
+        class _Fused(Stepper):
+            pingpong = set(self.pingpong)
+
+            def make_steps(self_inner, fixed_parameters=None, **kw):
+                from pystella_amd.field import var
+                from pystella_amd.elementwise import ElementWiseMap
+                dtv = var("dt")
+                self_inner._unknowns = []
+                for key in self_inner.rhs_dict:
+                    ff, outer = _field_of(key)
+                    self_inner._unknowns.append((ff, outer))
+                self_inner.dof_names = {ff.name for ff, _ in
+                                        self_inner._unknowns}
+                steps = []
+                for stage in range(self_inner.num_stages):
+                    tmp = {}
+                    rk = {}
+                    for i, (key, rhs_expr) in enumerate(
+                            self_inner.rhs_dict.items()):
+                        ff, outer = _field_of(key)
+                        k = Field(f"{ff.name}_tmp", offset=0,
+                                  shape=ff.shape, indices=ff.indices)
+                        k_acc = k[outer] if outer else k
+                        rhs_name = var(f"rhs_{i}")
+                        tmp[rhs_name] = rhs_expr
+                        rk[k_acc] = (self_inner._A[stage] * k_acc
+                                     + dtv * rhs_name)
+                        if ff.name in _Fused.pingpong:
+                            out_f = Field(f"{ff.name}_next",
+                                          offset=ff.offset,
+                                          shape=ff.shape,
+                                          indices=ff.indices)
+                            out_acc = out_f[outer] if outer else out_f
+                        else:
+                            out_acc = key
+                        rk[out_acc] = key + self_inner._B[stage] * k_acc
+                    steps.append(ElementWiseMap(
+                        rk, tmp_instructions=tmp,
+                        halo_shape=self_inner.halo_shape,
+                        rank_shape=self_inner.rank_shape,
+                        name=f"rk_stencil_stage{stage}",
+                        fixed_parameters=fixed_parameters, **kw))
+                self_inner.tmp_arrays = {}
+                return steps
+
+        self._stepper = _Fused(new_rhs, dt=dt, halo_shape=halo_shape,
+                               rank_shape=rank_shape, **kwargs)
+        self.num_stages = self._stepper.num_stages
+        self.expected_order = self._stepper.expected_order
+
+    def __call__(self, stage, queue=None, **kwargs):
+        self._stepper(stage, **kwargs)
+
+    @property
+    def tmp_arrays(self):
+        return self._stepper.tmp_arrays

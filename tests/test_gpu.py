@@ -233,3 +233,63 @@ def test_fused_lap_reduction_vs_unfused(grid_shape=(32, 32, 32), h=2):
             k, out_u[k], out_f[k])
     assert abs(out_u["total"] - out_f["total"]) < 1e-11 * abs(
         out_u["total"])
+
+
+@requires_gpu
+def test_stencil_stepper_vs_unfused_gpu(grid_shape=(24, 24, 24), h=2):
+    """Fully fused hot loop (inline-Laplacian stage kernels + no-lap
+    energy reduction) reproduces the reference-structure loop."""
+    from pystella_amd.fusion import (
+        FusedLaplacianReduction, StencilRKStepper)
+    from pystella_amd.sectors import get_rho_and_p
+
+    def potential(f):
+        return f[0]**2 / 2 + f[0]**2 * f[1]**2 / 4
+
+    sector = ps.ScalarSector(2, potential=potential)
+    decomp = ps.DomainDecomposition((1, 1, 1), h, rank_shape=grid_shape)
+    dx = (0.2, 0.2, 0.2)
+    dt = 0.01
+    derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=grid_shape)
+    gs = float(np.prod(grid_shape))
+    pad = tuple(n + 2 * h for n in grid_shape)
+    torch.manual_seed(11)
+    f0 = torch.rand((2,) + pad, dtype=torch.float64)
+    d0 = torch.rand((2,) + pad, dtype=torch.float64)
+    a = np.ones(1)
+    hub = 0.1 * np.ones(1)
+    cut = (slice(None),) + (slice(h, -h),) * 3
+
+    # reference-structure on GPU
+    fu, du = f0.clone().cuda(), d0.clone().cuda()
+    lap = torch.zeros((2,) + grid_shape, dtype=torch.float64,
+                      device="cuda")
+    st = ps.LowStorageRK54([sector], dt=dt, halo_shape=h,
+                           rank_shape=grid_shape)
+    red_u = FusedLaplacianReduction(decomp, sector, derivs, halo_shape=h,
+                                    grid_size=gs, callback=get_rho_and_p,
+                                    store_lap=True)
+    e_u = red_u(f=fu, dfdt=du, lap_f=lap, a=a)
+    for s in range(st.num_stages):
+        st(s, a=a, hubble=hub, f=fu, dfdt=du, lap_f=lap)
+        e_u = red_u(f=fu, dfdt=du, lap_f=lap, a=a)
+
+    # fused structure on GPU
+    ff, df = f0.clone().cuda(), d0.clone().cuda()
+    fnx = torch.zeros_like(ff)
+    fst = StencilRKStepper(ps.LowStorageRK54, [sector], derivs,
+                           halo_shape=h, rank_shape=grid_shape, dt=dt)
+    red_f = FusedLaplacianReduction(decomp, sector, derivs, halo_shape=h,
+                                    grid_size=gs, callback=get_rho_and_p,
+                                    store_lap=False)
+    arrays = {"f": ff, "dfdt": df, "f_next": fnx}
+    e_f = red_f(f=arrays["f"], dfdt=arrays["dfdt"], a=a)
+    for s in range(fst.num_stages):
+        fst(s, a=a, hubble=hub, **arrays)
+        arrays["f"], arrays["f_next"] = arrays["f_next"], arrays["f"]
+        e_f = red_f(f=arrays["f"], dfdt=arrays["dfdt"], a=a)
+    torch.cuda.synchronize()
+
+    assert (arrays["f"][cut] - fu[cut]).abs().max().item() < 1e-12
+    assert (arrays["dfdt"][cut] - du[cut]).abs().max().item() < 1e-12
+    assert abs(e_f["total"] - e_u["total"]) < 1e-10 * abs(e_u["total"])
