@@ -43,6 +43,10 @@ def main():
     parser.add_argument("--no-fuse", action="store_true",
                         help="reference-structure loop (separate lap "
                              "array + unfused stage kernels)")
+    parser.add_argument("--no-fuse-energy", action="store_true",
+                        help="keep the energy reduction as a separate "
+                             "fused lap+reduce kernel instead of folding "
+                             "it into the RK stage kernel")
     p = parser.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -80,13 +84,23 @@ def main():
     derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=rank_shape)
     from pystella_amd.fusion import (
         FusedLaplacianReduction, StencilRKStepper)
+    fuse_energy = not (p.no_fuse or p.no_fuse_energy)
     if p.no_fuse:
         stepper = ps.LowStorageRK54([sector], halo_shape=h,
                                     rank_shape=rank_shape, dt=dt)
+    elif fuse_energy:
+        # fully fused MI355X structure: one kernel per RK stage that
+        # evaluates the Laplacian inline (ping-pong f), updates the
+        # unknowns AND reduces the input-state energy — no separate
+        # energy pass in the hot loop at all
+        stepper = StencilRKStepper(ps.LowStorageRK54, [sector], derivs,
+                                   halo_shape=h, rank_shape=rank_shape,
+                                   dt=dt, reducers=sector,
+                                   grid_size=grid_size,
+                                   callback=get_rho_and_p)
     else:
-        # MI355X-optimized structure: stage kernels evaluate the
-        # Laplacian inline (ping-pong f); the energy reduction never
-        # materializes lap in HBM
+        # stage kernels evaluate the Laplacian inline (ping-pong f);
+        # the energy reduction is a separate fused lap+reduce kernel
         stepper = StencilRKStepper(ps.LowStorageRK54, [sector], derivs,
                                    halo_shape=h, rank_shape=rank_shape,
                                    dt=dt)
@@ -123,11 +137,23 @@ def main():
     def step():
         nonlocal energy
         for s in range(stepper.num_stages):
-            stepper(s, a=expand.a, hubble=expand.hubble, **arrays)
-            if not p.no_fuse:
-                arrays["f"], arrays["f_next"] =                     arrays["f_next"], arrays["f"]
-            expand.step(s, energy["total"], energy["pressure"], dt)
-            energy = compute_energy(expand.a)
+            if fuse_energy:
+                # the stage kernel itself returns the input-state
+                # energy — identical values to the reference loop's
+                # standalone reduction after the previous stage
+                energy = stepper(s, a=expand.a, hubble=expand.hubble,
+                                 **arrays)
+                arrays["f"], arrays["f_next"] = \
+                    arrays["f_next"], arrays["f"]
+                decomp.share_halos(arrays["f"])
+                expand.step(s, energy["total"], energy["pressure"], dt)
+            else:
+                stepper(s, a=expand.a, hubble=expand.hubble, **arrays)
+                if not p.no_fuse:
+                    arrays["f"], arrays["f_next"] = \
+                        arrays["f_next"], arrays["f"]
+                expand.step(s, energy["total"], energy["pressure"], dt)
+                energy = compute_energy(expand.a)
 
     def sync():
         if on_gpu:

@@ -154,6 +154,117 @@ def get_elementwise_kernel(map_dict, tmp_instructions, field_args,
 
 
 # ---------------------------------------------------------------------------
+# JIT'd fused elementwise map + simultaneous reductions over the INPUT
+# state (MI355X traffic optimization: the RK stage kernel reads f, dfdt
+# and computes lap f inline anyway — accumulating the energy reducers in
+# the same pass removes the standalone lap+reduction kernel from the hot
+# loop entirely; see fusion.StencilRKStepper).
+
+STAGERED_TEMPLATE = """{defines}
+{preamble}
+#define NRED {nred}
+extern "C" __global__ __launch_bounds__(TBZ * TBY) void {name}(
+    {params})
+{{
+    double acc[NRED];
+    {init}
+    const int k = blockIdx.x * TBZ + (threadIdx.x % TBZ);
+    const int j = blockIdx.y * TBY + (threadIdx.x / TBZ);
+    const int i0 = blockIdx.z * XCHUNK;
+    const int i1 = (i0 + XCHUNK < NX) ? i0 + XCHUNK : NX;
+    if (k < NZ && j < NY) {{
+        for (int i = i0; i < i1; ++i) {{
+            {body}
+        }}
+    }}
+"""
+
+
+class JitStageReduction:
+    """Fused per-site map + multi-quantity reduction of the pre-update
+    state.  Emission order per site: temporaries (which include the
+    inline Laplacian), reduction accumulation (reads input values), then
+    the update stores — so the reducers see the input state even for
+    in-place unknowns."""
+
+    def __init__(self, map_dict, tmp_instructions, entries, field_args,
+                 scalar_names, halo, rank_shape, name="rk_stage_red",
+                 tile=(64, 4, 64)):
+        self.rank_shape = tuple(rank_shape)
+        self.tile = tile
+        self.entries = entries
+        self.field_args = [fa for fa in field_args if fa.spatial]
+        nred = len(entries)
+        cg = Codegen(field_args, halo, rank_shape)
+
+        lines = []
+        for lhs, rhs in (tmp_instructions or {}).items():
+            tname = lhs.name if hasattr(lhs, "name") else str(lhs)
+            cg.tmp_names.add(tname)
+            lines.append(f"const double {tname} = {cg.emit(rhs)};")
+        init_lines, combine_cases = [], []
+        for r, (expr, op) in enumerate(entries):
+            init_lines.append(f"acc[{r}] = {_OP_INIT[op]};")
+            comb = _OP_COMBINE[op]
+            val = cg.emit(expr)
+            lines.append(
+                "{ const double a = acc[%d]; const double b = %s; "
+                "acc[%d] = %s; }" % (r, val, r, comb))
+            combine_cases.append(f"(r == {r}) ? {comb} : ")
+        for lhs, rhs in map_dict.items():
+            lines.append(f"{cg.emit(lhs)} = {cg.emit(rhs)};")
+        combine = "".join(combine_cases) + "0.0"
+
+        ptr_params = ", ".join(
+            f"double* __restrict__ {fa.name}" for fa in self.field_args)
+        dbl_params = ", ".join(f"double {c}" for c, _ in cg.scalars)
+        params = ", ".join(x for x in (
+            ptr_params, "double* __restrict__ partials", dbl_params) if x)
+
+        defines = geometry_defines(halo, rank_shape)
+        defines += _tile_defines(tile, rank_shape)
+        defines += "#define COMBINE(r, a, b) (" + combine + ")\n"
+        src = (STAGERED_TEMPLATE + REDUCTION_TAIL).format(
+            defines=defines, preamble=PREAMBLE, nred=nred, name=name,
+            params=params, init="\n    ".join(init_lines),
+            body="\n            ".join(lines))
+        self.source = src
+        self.scalar_keys = [k for _, k in cg.scalars]
+        self.key = ext().jit_compile(src, name)
+        self.grid = _tile_grid(tile, rank_shape)
+        self.block = tile[0] * tile[1]
+        self.nblk = self.grid[0] * self.grid[1] * self.grid[2]
+        self._partials = None
+
+    def __call__(self, env):
+        dev = None
+        ptrs = []
+        for fa in self.field_args:
+            t = _check_tensor(fa.name, env[fa.name])
+            dev = t.device
+            ptrs.append(t.data_ptr())
+        nred = len(self.entries)
+        if (self._partials is None
+                or self._partials.device != dev
+                or self._partials.shape[1] != self.nblk):
+            self._partials = torch.empty((nred, self.nblk),
+                                         dtype=torch.float64, device=dev)
+        doubles = [_resolve_scalar(env, k) for k in self.scalar_keys]
+        ext().jit_launch(self.key, self.grid[0], self.grid[1],
+                         self.grid[2], self.block, 1, 1, 0, _stream(),
+                         ptrs + [self._partials.data_ptr()], [], doubles)
+        return self._finish(dev)
+
+
+def get_stage_reduction_kernel(map_dict, tmp_instructions, entries,
+                               field_args, scalar_names, halo, rank_shape,
+                               name="rk_stage_red", tile=(64, 4, 64)):
+    return JitStageReduction(map_dict, tmp_instructions, entries,
+                             field_args, scalar_names, halo, rank_shape,
+                             name=name, tile=tile)
+
+
+# ---------------------------------------------------------------------------
 # JIT'd simultaneous reductions
 
 REDUCTION_TAIL = """
@@ -285,6 +396,9 @@ class JitReduction:
                          self.grid[2], self.block, 1, 1, 0, _stream(),
                          ptrs + [self._partials.data_ptr()], [], doubles)
         return self._finish(dev)
+
+
+JitStageReduction._finish = JitReduction._finish
 
 
 def get_reduction_kernel(entries, field_args, scalar_names, halo,

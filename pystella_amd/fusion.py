@@ -81,6 +81,87 @@ class FusedLaplacianReduction(Reduction):
         return self._combine(local, rank_shape)
 
 
+class _StageRedMap:
+    """One RK stage kernel fused with input-state reductions.
+
+    GPU: a single JIT kernel (``backend.hip.JitStageReduction``) whose
+    per-site order is temporaries (incl. the inline Laplacian) →
+    reduction accumulation → update stores, so the reducers see the
+    stage's input state.  CPU oracle: the unfused compose — Laplacian
+    scratch via the FiniteDifferencer, torch reduction, then the stage
+    map.  Returns ``(local_reduction_values, rank_shape)``.
+    """
+
+    def __init__(self, map_dict, tmp_instructions=None, red_entries=(),
+                 reduction=None, derivs=None, lap_names=(), **kwargs):
+        from pystella_amd.elementwise import ElementWiseMap
+        self._map = ElementWiseMap(map_dict, tmp_instructions, **kwargs)
+        self.red_entries = list(red_entries)
+        self.reduction = reduction
+        self.derivs = derivs
+        self.lap_names = list(lap_names)
+        # extend argument discovery with the reducer expressions
+        from pystella_amd.field import (
+            Field, Variable, collect_fields, get_field_args, iter_exprs,
+            walk_expr)
+        exprs = [e for e, _ in self.red_entries]
+        everything = [self._map.tmp_instructions, self._map.map_dict,
+                      exprs]
+        self._map.fields = collect_fields(everything)
+        self._map.field_args = get_field_args(everything)
+        self._map.arg_names = {f.name for f in self._map.fields}
+        tmp_names = {k.name for k in self._map.tmp_instructions}
+        scal = set(self._map.scalar_names)
+
+        def visit(x):
+            if isinstance(x, Variable) and not isinstance(x, Field):
+                if x.name not in tmp_names:
+                    scal.add(x.name)
+
+        for e in iter_exprs(exprs):
+            walk_expr(e, visit)
+        self._map.scalar_names = scal
+        self._hip_kernel = None
+
+    def __call__(self, queue=None, **kwargs):
+        import torch as _torch
+        m = self._map
+        env = m._build_env(kwargs)
+        rank_shape = m._infer_rank_shape(env)
+        on_gpu = any(isinstance(v, _torch.Tensor) and v.is_cuda
+                     for v in env.values())
+        if on_gpu:
+            if self._hip_kernel is None or \
+                    self._hip_kernel.rank_shape != rank_shape:
+                from pystella_amd.backend.hip import (
+                    get_stage_reduction_kernel)
+                self._hip_kernel = get_stage_reduction_kernel(
+                    m.map_dict, m.tmp_instructions, self.red_entries,
+                    m.field_args, sorted(m.scalar_names), m.halo_shape,
+                    rank_shape, name=m.name)
+            local = self._hip_kernel(env)
+            return local, rank_shape
+        # CPU oracle: reduction of the input state with a lap scratch,
+        # then the stage update
+        from itertools import product
+        env2 = dict(env)
+        for lap_name in self.lap_names:
+            f_name = lap_name[len("lap_"):]
+            f = env[f_name]
+            shape = f.shape[:-3] + tuple(rank_shape)
+            lap = torch.zeros(shape, dtype=f.dtype, device=f.device)
+            for s in product(*[range(n) for n in f.shape[:-3]]):
+                self.derivs._apply_lap_cpu(f[s], lap[s])
+            env2[lap_name] = lap
+        local = self.reduction._local_torch(env2, rank_shape)
+        from pystella_amd.backend.torcheval import (
+            EvalContext, eval_statements)
+        ctx = EvalContext(m.halo_shape, rank_shape)
+        eval_statements(m.map_dict, env, ctx,
+                        tmp_statements=m.tmp_instructions)
+        return local, rank_shape
+
+
 class StencilRKStepper:
     """Low-storage RK stepper whose stage kernels evaluate the Laplacian
     *inline* from the finite-difference stencil instead of reading a
@@ -107,9 +188,11 @@ class StencilRKStepper:
     """
 
     def __init__(self, Stepper, input, derivs, halo_shape=0,
-                 rank_shape=None, dt=None, **kwargs):
+                 rank_shape=None, dt=None, reducers=None, grid_size=None,
+                 callback=None, **kwargs):
         from pystella_amd.field import (
-            DynamicField, Field, Subscript, substitute, collect_fields)
+            DynamicField, Field, Subscript, substitute, collect_fields,
+            var)
         from pystella_amd.derivs import _LAP_COEFS, centered_diff
         from pystella_amd.step import LowStorageRKStepper, _field_of
         from pystella_amd.sectors import Sector
@@ -130,12 +213,24 @@ class StencilRKStepper:
         dx = derivs.dx
         coefs = _LAP_COEFS[h]
 
-        # find DynamicFields whose .lap is referenced; build inline
-        # stencil substitutions and the ping-pong name map
+        # find DynamicFields whose .lap is referenced; compute each
+        # Laplacian component once per site into a named temporary, and
+        # substitute that temporary for lap accesses in the equations of
+        # motion AND (below) the energy reducers; build the ping-pong
+        # name map
         fields = collect_fields(list(rhs_dict.values()))
+        if reducers is not None:
+            red_input = (reducers.reducers
+                         if isinstance(reducers, Sector) else reducers)
+            red_exprs = []
+            for v in red_input.values():
+                red_exprs.extend(v if isinstance(v, list) else [v])
+            fields |= collect_fields(
+                [e[0] if isinstance(e, tuple) else e for e in red_exprs])
         lap_names = {f.name for f in fields if f.name.startswith("lap_")}
         self.pingpong = []
         subs = {}
+        lap_tmps = {}
         for key in rhs_dict:
             f, outer = _field_of(key)
             lap_name = f"lap_{f.name}"
@@ -150,11 +245,31 @@ class StencilRKStepper:
                                       order=2) * (1.0 / dx[mu] ** 2)
                         for mu in range(3))
                     lap_acc = lap_f[fld] if f.shape else lap_f
-                    subs[lap_acc] = lap_expr
+                    lv = var(f"lapv_{f.name}_{fld}")
+                    lap_tmps[lv] = lap_expr
+                    subs[lap_acc] = lv
         self.pingpong = sorted(set(self.pingpong))
         new_rhs = {k: substitute(v, subs) for k, v in rhs_dict.items()}
 
+        # fused input-state reducers (energy each RK stage without a
+        # separate lap+reduction pass)
+        self._reduction = None
+        red_entries = []
+        if reducers is not None:
+            self._reduction = Reduction(
+                derivs.decomp, reducers, halo_shape=halo_shape,
+                rank_shape=rank_shape, grid_size=grid_size,
+                callback=callback)
+            red_entries = [(substitute(expr, subs), op)
+                           for _, _, expr, op in self._reduction.flat]
+        self._derivs = derivs
+        self._lap_names = sorted(lap_names)
+
         # redirect writes of ping-ponged fields to NAME_next
+        reduction = self._reduction
+        lap_name_list = self._lap_names
+        derivs_ref = derivs
+
         class _Fused(Stepper):
             pingpong = set(self.pingpong)
 
@@ -170,7 +285,7 @@ class StencilRKStepper:
                                         self_inner._unknowns}
                 steps = []
                 for stage in range(self_inner.num_stages):
-                    tmp = {}
+                    tmp = dict(lap_tmps)
                     rk = {}
                     for i, (key, rhs_expr) in enumerate(
                             self_inner.rhs_dict.items()):
@@ -191,12 +306,23 @@ class StencilRKStepper:
                         else:
                             out_acc = key
                         rk[out_acc] = key + self_inner._B[stage] * k_acc
-                    steps.append(ElementWiseMap(
-                        rk, tmp_instructions=tmp,
-                        halo_shape=self_inner.halo_shape,
-                        rank_shape=self_inner.rank_shape,
-                        name=f"rk_stencil_stage{stage}",
-                        fixed_parameters=fixed_parameters, **kw))
+                    if reduction is not None:
+                        steps.append(_StageRedMap(
+                            rk, tmp_instructions=tmp,
+                            red_entries=red_entries,
+                            reduction=reduction, derivs=derivs_ref,
+                            lap_names=lap_name_list,
+                            halo_shape=self_inner.halo_shape,
+                            rank_shape=self_inner.rank_shape,
+                            name=f"rk_stage_red{stage}",
+                            fixed_parameters=fixed_parameters, **kw))
+                    else:
+                        steps.append(ElementWiseMap(
+                            rk, tmp_instructions=tmp,
+                            halo_shape=self_inner.halo_shape,
+                            rank_shape=self_inner.rank_shape,
+                            name=f"rk_stencil_stage{stage}",
+                            fixed_parameters=fixed_parameters, **kw))
                 self_inner.tmp_arrays = {}
                 return steps
 
@@ -206,7 +332,15 @@ class StencilRKStepper:
         self.expected_order = self._stepper.expected_order
 
     def __call__(self, stage, queue=None, **kwargs):
-        self._stepper(stage, **kwargs)
+        """Runs stage ``stage``; when ``reducers`` were given, returns
+        the reduced quantities of the stage's INPUT state (the same
+        values the reference loop obtains from its standalone energy
+        reduction after the previous stage)."""
+        result = self._stepper(stage, **kwargs)
+        if self._reduction is not None and result is not None:
+            local, rank_shape = result
+            return self._reduction._combine(local, rank_shape)
+        return result
 
     @property
     def tmp_arrays(self):

@@ -287,7 +287,7 @@ class LowStorageRKStepper(Stepper):
     def __call__(self, stage, queue=None, filter_args=None, **kwargs):
         if not self.tmp_arrays:
             self.tmp_arrays = self.get_tmp_arrays_like(**kwargs)
-        self.steps[stage](**kwargs, **self.tmp_arrays)
+        return self.steps[stage](**kwargs, **self.tmp_arrays)
 
 
 class LowStorageRK54(LowStorageRKStepper):

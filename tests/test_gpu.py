@@ -293,3 +293,57 @@ def test_stencil_stepper_vs_unfused_gpu(grid_shape=(24, 24, 24), h=2):
     assert (arrays["f"][cut] - fu[cut]).abs().max().item() < 1e-12
     assert (arrays["dfdt"][cut] - du[cut]).abs().max().item() < 1e-12
     assert abs(e_f["total"] - e_u["total"]) < 1e-10 * abs(e_u["total"])
+
+
+@requires_gpu
+def test_stage_fused_energy_gpu(grid_shape=(32, 32, 32)):
+    """GPU energy-fused RK stage kernel vs the CPU oracle: same energy
+    values and same updated fields."""
+    from pystella_amd.fusion import StencilRKStepper
+    from pystella_amd.sectors import get_rho_and_p
+    h = 2
+    decomp = ps.DomainDecomposition((1, 1, 1), h, rank_shape=grid_shape)
+    dx = (0.3, 0.31, 0.32)
+    dt = 0.01
+    gsize = float(np.prod(grid_shape))
+
+    def pot(f):
+        return f[0]**2 / 2 + f[0]**2 * f[1]**2 / 4
+
+    sector = ps.ScalarSector(2, potential=pot)
+    derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=grid_shape)
+    pad = tuple(n + 2 * h for n in grid_shape)
+    torch.manual_seed(5)
+    f0 = torch.rand((2,) + pad, dtype=torch.float64)
+    d0 = torch.rand((2,) + pad, dtype=torch.float64)
+    a = np.ones(1)
+    hub = 0.1 * np.ones(1)
+    cut = (slice(None),) + (slice(h, -h),) * 3
+
+    def run(device):
+        ff = f0.clone().to(device)
+        df = d0.clone().to(device)
+        fst = StencilRKStepper(ps.LowStorageRK54, [sector], derivs,
+                               halo_shape=h, rank_shape=grid_shape,
+                               dt=dt, reducers=sector, grid_size=gsize,
+                               callback=get_rho_and_p)
+        arrays = {"f": ff, "dfdt": df, "f_next": torch.zeros_like(ff)}
+        decomp.share_halos(arrays["f"])
+        energies = []
+        for s in range(fst.num_stages):
+            energies.append(fst(s, a=a, hubble=hub, **arrays))
+            arrays["f"], arrays["f_next"] = \
+                arrays["f_next"], arrays["f"]
+            decomp.share_halos(arrays["f"])
+        return arrays, energies
+
+    arr_c, en_c = run("cpu")
+    arr_g, en_g = run("cuda")
+    torch.cuda.synchronize()
+    for s, (ec, eg) in enumerate(zip(en_c, en_g)):
+        for key in ec:
+            assert np.allclose(np.asarray(ec[key]), np.asarray(eg[key]),
+                               rtol=1e-12), (s, key)
+    assert (arr_g["f"].cpu()[cut] - arr_c["f"][cut]).abs().max() < 1e-12
+    assert (arr_g["dfdt"].cpu()[cut]
+            - arr_c["dfdt"][cut]).abs().max() < 1e-12

@@ -171,3 +171,63 @@ def test_stencil_stepper_cpu_equivalence(grid_shape=(12, 12, 12)):
 
     assert (arrays["f"][cut] - fu[cut]).abs().max().item() < 1e-15
     assert (arrays["dfdt"][cut] - du[cut]).abs().max().item() < 1e-15
+
+
+def test_stage_fused_energy_cpu(grid_shape=(12, 12, 12)):
+    """The energy-fused stage kernel returns exactly the reduction
+    values of its input state and applies the same update."""
+    from pystella_amd.fusion import StencilRKStepper
+    from pystella_amd.sectors import get_rho_and_p
+    h = 2
+    decomp = ps.DomainDecomposition((1, 1, 1), h, rank_shape=grid_shape)
+    dx = (0.3, 0.3, 0.3)
+    dt = 0.01
+    gsize = float(np.prod(grid_shape))
+
+    def pot(f):
+        return f[0]**2 / 2 + f[0]**2 * f[1]**2 / 4
+
+    sector = ps.ScalarSector(2, potential=pot)
+    derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=grid_shape)
+    pad = tuple(n + 2 * h for n in grid_shape)
+    torch.manual_seed(3)
+    f0 = torch.rand((2,) + pad, dtype=torch.float64)
+    d0 = torch.rand((2,) + pad, dtype=torch.float64)
+    a = np.ones(1)
+    hub = 0.1 * np.ones(1)
+    cut = (slice(None),) + (slice(h, -h),) * 3
+
+    # reference: unfused loop + standalone reduction of the input state
+    red = ps.Reduction(decomp, sector, halo_shape=h,
+                       callback=get_rho_and_p, rank_shape=grid_shape,
+                       grid_size=gsize)
+    fu, du = f0.clone(), d0.clone()
+    lap = torch.zeros((2,) + grid_shape, dtype=torch.float64)
+    st = ps.LowStorageRK54([sector], dt=dt, halo_shape=h,
+                           rank_shape=grid_shape)
+    ref_energies = []
+    for s in range(st.num_stages):
+        derivs(fx=fu, lap=lap)
+        ref_energies.append(red(f=fu, dfdt=du, lap_f=lap, a=a))
+        st(s, a=a, hubble=hub, f=fu, dfdt=du, lap_f=lap)
+
+    # fused: stage kernel returns the input-state energy
+    ff, df = f0.clone(), d0.clone()
+    fst = StencilRKStepper(ps.LowStorageRK54, [sector], derivs,
+                           halo_shape=h, rank_shape=grid_shape, dt=dt,
+                           reducers=sector, grid_size=gsize,
+                           callback=get_rho_and_p)
+    arrays = {"f": ff, "dfdt": df, "f_next": torch.zeros_like(ff)}
+    decomp.share_halos(arrays["f"])
+    for s in range(fst.num_stages):
+        e_in = fst(s, a=a, hubble=hub, **arrays)
+        arrays["f"], arrays["f_next"] = arrays["f_next"], arrays["f"]
+        decomp.share_halos(arrays["f"])
+        for key in ("kinetic", "potential", "gradient"):
+            assert np.allclose(e_in[key], ref_energies[s][key],
+                               rtol=1e-13), (s, key)
+        assert np.allclose(e_in["total"], ref_energies[s]["total"],
+                           rtol=1e-13)
+
+    assert (arrays["f"][cut] - fu[cut]).abs().max().item() < 1e-14
+    assert (arrays["dfdt"][cut] - du[cut]).abs().max().item() < 1e-14

@@ -105,6 +105,38 @@ def main(n=512):
     results.sort()
     print("  BEST:", results[0])
 
+    # --- energy-fused RK stage kernel sweep (the current hot-loop
+    # kernel: update + input-state energy reduction in one pass)
+    from pystella_amd.fusion import StencilRKStepper
+    derivs0 = ps.FiniteDifferencer(decomp, h, dx, rank_shape=grid)
+    fst = StencilRKStepper(ps.LowStorageRK54, [sector], derivs0,
+                           halo_shape=h, rank_shape=grid, dt=dt,
+                           reducers=sector, grid_size=sites,
+                           callback=get_rho_and_p)
+    sm = fst._stepper.steps[1]._map
+    env3 = dict(a=a, hubble=hub, f=f, f_next=torch.zeros_like(f),
+                dfdt=dfdt,
+                f_tmp=torch.zeros((nscalars,) + grid,
+                                  dtype=torch.float64, device=device),
+                dfdt_tmp=torch.zeros((nscalars,) + grid,
+                                     dtype=torch.float64, device=device),
+                dt=dt)
+    red_entries = fst._stepper.steps[1].red_entries
+    print(f"== energy-fused rk stage kernel (16 passes ~ "
+          f"{16*GBms:.1f} GB)")
+    results = []
+    for tile in tiles:
+        k = H.JitStageReduction(
+            sm.map_dict, sm.tmp_instructions, red_entries,
+            sm.field_args, [], (h,) * 3, grid,
+            name=f"tune_sr_{tile[0]}_{tile[1]}_{tile[2]}", tile=tile)
+        ms = timeit(lambda: k(env3))
+        bw = 16 * GBms / ms / nscalars
+        results.append((ms, tile))
+        print(f"  tile={tile}:  {ms:7.3f} ms   {bw:5.2f} TB/s")
+    results.sort()
+    print("  BEST:", results[0])
+
     # --- AOT gradlap XCHUNK sweep (env var)
     derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=grid)
     print(f"== AOT gradlap lap-only (4 passes ~ {4*GBms:.1f} GB)")
