@@ -740,6 +740,147 @@ class JitLapReduction:
         return self._finish(dev)
 
 
+LAPSTAGE_TEMPLATE = """{defines}
+{preamble}
+#define NRED {nred}
+#define NF {nf}
+extern "C" __global__ __launch_bounds__(TBZ * TBY) void {name}(
+    {params})
+{{
+    double acc[NRED];
+    {init}
+    const int k = blockIdx.x * TBZ + (threadIdx.x % TBZ);
+    const int j = blockIdx.y * TBY + (threadIdx.x / TBZ);
+    const int i0 = blockIdx.z * XCHUNK;
+    const int i1 = (i0 + XCHUNK < NX) ? i0 + XCHUNK : NX;
+    if (k < NZ && j < NY) {{
+        const long sx = PSY * PSZ;
+        double ring[NF][2 * H + 1];
+        #pragma unroll
+        for (int fld = 0; fld < NF; ++fld) {{
+            const double* fp = {fname} + (long)fld * PVOL
+                               + (long)(j + H) * PSZ + (k + H);
+            #pragma unroll
+            for (int p = 0; p < 2 * H; ++p)
+                ring[fld][p] = fp[(long)(i0 + p) * sx];
+        }}
+        for (int i = i0; i < i1; ++i) {{
+            double lapv[NF];
+            #pragma unroll
+            for (int fld = 0; fld < NF; ++fld) {{
+                const double* fp = {fname} + (long)fld * PVOL
+                                   + (long)(j + H) * PSZ + (k + H);
+                ring[fld][2 * H] = fp[(long)(i + 2 * H) * sx];
+                const double* cp = fp + (long)(i + H) * sx;
+                double la = ring[fld][H] * LAPC0;
+                {lap_terms}
+                lapv[fld] = la;
+            }}
+            {body}
+            #pragma unroll
+            for (int fld = 0; fld < NF; ++fld)
+                #pragma unroll
+                for (int p = 0; p < 2 * H; ++p)
+                    ring[fld][p] = ring[fld][p + 1];
+        }}
+    }}
+""" + REDUCTION_TAIL
+
+
+class JitLapStage:
+    """Ring-buffer RK stage kernel fused with input-state reductions:
+    the stencil field marches along x through a per-thread register ring
+    (one new load per site instead of 4h+1), the Laplacian is formed in
+    registers, the reducers accumulate input-state values, and the
+    2N-storage update stores last.  This is the fastest form of the hot
+    loop's only kernel (see fusion.StencilRKStepper)."""
+
+    def __init__(self, map_dict, tmp_instructions, entries, field_args,
+                 scalar_names, halo, rank_shape, dx, nf, f_name="f",
+                 lap_name="lap_f", name="rk_lapstage", tile=(64, 4, 32)):
+        from pystella_amd.derivs import _LAP_COEFS
+        self.rank_shape = tuple(rank_shape)
+        self.tile = tile
+        self.entries = entries
+        h = max(halo) if isinstance(halo, (tuple, list)) else halo
+        self.nf = nf
+        cg = _LapCodegen(field_args, halo, rank_shape, f_name, lap_name)
+
+        lines = []
+        for lhs, rhs in (tmp_instructions or {}).items():
+            tname = lhs.name if hasattr(lhs, "name") else str(lhs)
+            cg.tmp_names.add(tname)
+            lines.append(f"const double {tname} = {cg.emit(rhs)};")
+        init_lines, combine_cases = [], []
+        for r, (expr, op) in enumerate(entries):
+            init_lines.append(f"acc[{r}] = {_OP_INIT[op]};")
+            comb = _OP_COMBINE[op]
+            val = cg.emit(expr)
+            lines.append(
+                "{ const double a = acc[%d]; const double b = %s; "
+                "acc[%d] = %s; }" % (r, val, r, comb))
+            combine_cases.append(f"(r == {r}) ? {comb} : ")
+        for lhs, rhs in map_dict.items():
+            lines.append(f"{cg.emit(lhs)} = {cg.emit(rhs)};")
+        combine = "".join(combine_cases) + "0.0"
+
+        inv2 = [1.0 / d / d for d in dx]
+        coefs = _LAP_COEFS[h]
+        lap_terms = []
+        for s in range(1, h + 1):
+            c = coefs[s]
+            lap_terms.append(
+                f"la += {c!r} * ((ring[fld][H+{s}] + ring[fld][H-{s}])"
+                f"*{inv2[0]!r} + (cp[{s}*PSZ] + cp[-{s}*PSZ])*{inv2[1]!r}"
+                f" + (cp[{s}] + cp[-{s}])*{inv2[2]!r});")
+        lapc0 = coefs[0] * (inv2[0] + inv2[1] + inv2[2])
+
+        # pointer params: stencil field first, then every other spatial
+        # field referenced by the statements or reducers
+        self.ptr_names = [f_name] + sorted(
+            fa.name for fa in field_args
+            if fa.spatial and fa.name not in (f_name, lap_name))
+        by_name = {fa.name: fa for fa in field_args}
+        self.field_args = [by_name[n] for n in self.ptr_names
+                           if n in by_name]
+        ptr_params = ", ".join(
+            f"double* __restrict__ {n}" for n in self.ptr_names)
+        dbl_params = ", ".join(f"double {c}" for c, _ in cg.scalars)
+        params = ", ".join(x for x in (
+            ptr_params, "double* __restrict__ partials", dbl_params) if x)
+
+        defines = geometry_defines(halo, rank_shape)
+        defines += _tile_defines(tile, rank_shape)
+        defines += f"#define COMBINE(r, a, b) ({combine})\n"
+        defines += f"#define LAPC0 ({lapc0!r})\n"
+        src = LAPSTAGE_TEMPLATE.format(
+            defines=defines, preamble=PREAMBLE, nred=len(entries), nf=nf,
+            name=name, params=params, fname=f_name,
+            init="\n    ".join(init_lines),
+            lap_terms="\n                ".join(lap_terms),
+            body="\n            ".join(lines))
+        self.source = src
+        self.scalar_keys = [k for _, k in cg.scalars]
+        self.key = ext().jit_compile(src, name)
+        self.grid = _tile_grid(tile, rank_shape)
+        self.block = tile[0] * tile[1]
+        self.nblk = self.grid[0] * self.grid[1] * self.grid[2]
+        self._partials = None
+
+    _finish = JitReduction._finish
+    __call__ = JitLapReduction.__call__
+
+
+def get_lap_stage_kernel(map_dict, tmp_instructions, entries, field_args,
+                         scalar_names, halo, rank_shape, dx, nf,
+                         f_name="f", lap_name="lap_f",
+                         name="rk_lapstage", tile=(64, 4, 32)):
+    return JitLapStage(map_dict, tmp_instructions, entries, field_args,
+                       scalar_names, halo, rank_shape, dx, nf,
+                       f_name=f_name, lap_name=lap_name, name=name,
+                       tile=tile)
+
+
 def get_lap_reduction_kernel(entries, field_args, scalar_names, halo,
                              rank_shape, dx, nf, f_name="f",
                              lap_name="lap_f", tile=(64, 4, 64),

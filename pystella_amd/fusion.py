@@ -93,13 +93,24 @@ class _StageRedMap:
     """
 
     def __init__(self, map_dict, tmp_instructions=None, red_entries=(),
-                 reduction=None, derivs=None, lap_names=(), **kwargs):
+                 reduction=None, derivs=None, lap_names=(),
+                 ring=None, **kwargs):
         from pystella_amd.elementwise import ElementWiseMap
         self._map = ElementWiseMap(map_dict, tmp_instructions, **kwargs)
         self.red_entries = list(red_entries)
         self.reduction = reduction
         self.derivs = derivs
         self.lap_names = list(lap_names)
+        # ring = (rk_orig, tmp_orig, red_entries_orig, f_name, nf):
+        # the pre-substitution statements (lap accesses intact) for the
+        # register-ring GPU kernel; available when exactly one stencil
+        # field is involved
+        self.ring = ring
+        if ring is not None:
+            from pystella_amd.field import get_field_args
+            rk_o, tmp_o, red_o, _, _ = ring
+            self._ring_field_args = get_field_args(
+                [tmp_o, rk_o, [e for e, _ in red_o]])
         # extend argument discovery with the reducer expressions
         from pystella_amd.field import (
             Field, Variable, collect_fields, get_field_args, iter_exprs,
@@ -133,12 +144,22 @@ class _StageRedMap:
         if on_gpu:
             if self._hip_kernel is None or \
                     self._hip_kernel.rank_shape != rank_shape:
-                from pystella_amd.backend.hip import (
-                    get_stage_reduction_kernel)
-                self._hip_kernel = get_stage_reduction_kernel(
-                    m.map_dict, m.tmp_instructions, self.red_entries,
-                    m.field_args, sorted(m.scalar_names), m.halo_shape,
-                    rank_shape, name=m.name)
+                if self.ring is not None:
+                    from pystella_amd.backend.hip import (
+                        get_lap_stage_kernel)
+                    rk_o, tmp_o, red_o, f_name, nf = self.ring
+                    self._hip_kernel = get_lap_stage_kernel(
+                        rk_o, tmp_o, red_o, self._ring_field_args, [],
+                        m.halo_shape, rank_shape, self.derivs.dx, nf,
+                        f_name=f_name, lap_name=f"lap_{f_name}",
+                        name=m.name)
+                else:
+                    from pystella_amd.backend.hip import (
+                        get_stage_reduction_kernel)
+                    self._hip_kernel = get_stage_reduction_kernel(
+                        m.map_dict, m.tmp_instructions, self.red_entries,
+                        m.field_args, sorted(m.scalar_names),
+                        m.halo_shape, rank_shape, name=m.name)
             local = self._hip_kernel(env)
             return local, rank_shape
         # CPU oracle: reduction of the input state with a lap scratch,
@@ -269,6 +290,19 @@ class StencilRKStepper:
         reduction = self._reduction
         lap_name_list = self._lap_names
         derivs_ref = derivs
+        rhs_dict_orig = dict(rhs_dict)
+        # register-ring kernel eligibility: a single stencil field
+        ring_f_name = None
+        ring_nf = 0
+        if reduction is not None and len(lap_name_list) == 1:
+            ring_f_name = lap_name_list[0][len("lap_"):]
+            for fld in fields:
+                if fld.name == ring_f_name:
+                    ring_nf = fld.shape[0] if fld.shape else 1
+                    break
+        red_entries_orig = ([(expr, op)
+                             for _, _, expr, op in reduction.flat]
+                            if reduction is not None else [])
 
         class _Fused(Stepper):
             pingpong = set(self.pingpong)
@@ -287,6 +321,8 @@ class StencilRKStepper:
                 for stage in range(self_inner.num_stages):
                     tmp = dict(lap_tmps)
                     rk = {}
+                    tmp_o = {}
+                    rk_o = {}
                     for i, (key, rhs_expr) in enumerate(
                             self_inner.rhs_dict.items()):
                         ff, outer = _field_of(key)
@@ -295,8 +331,10 @@ class StencilRKStepper:
                         k_acc = k[outer] if outer else k
                         rhs_name = var(f"rhs_{i}")
                         tmp[rhs_name] = rhs_expr
+                        tmp_o[rhs_name] = rhs_dict_orig[key]
                         rk[k_acc] = (self_inner._A[stage] * k_acc
                                      + dtv * rhs_name)
+                        rk_o[k_acc] = rk[k_acc]
                         if ff.name in _Fused.pingpong:
                             out_f = Field(f"{ff.name}_next",
                                           offset=ff.offset,
@@ -306,12 +344,17 @@ class StencilRKStepper:
                         else:
                             out_acc = key
                         rk[out_acc] = key + self_inner._B[stage] * k_acc
+                        rk_o[out_acc] = rk[out_acc]
                     if reduction is not None:
+                        ring = None
+                        if ring_f_name is not None:
+                            ring = (rk_o, tmp_o, red_entries_orig,
+                                    ring_f_name, ring_nf)
                         steps.append(_StageRedMap(
                             rk, tmp_instructions=tmp,
                             red_entries=red_entries,
                             reduction=reduction, derivs=derivs_ref,
-                            lap_names=lap_name_list,
+                            lap_names=lap_name_list, ring=ring,
                             halo_shape=self_inner.halo_shape,
                             rank_shape=self_inner.rank_shape,
                             name=f"rk_stage_red{stage}",

@@ -137,6 +137,26 @@ def main(n=512):
     results.sort()
     print("  BEST:", results[0])
 
+    # --- register-ring energy-fused stage kernel sweep (current
+    # default hot-loop kernel)
+    sm1 = fst._stepper.steps[1]
+    rk_o, tmp_o, red_o, f_name, nf = sm1.ring
+    print(f"== ring energy-fused rk stage kernel (16 passes ~ "
+          f"{16*GBms/nscalars:.1f} GB)")
+    results = []
+    for tile in tiles + [(128, 2, 32), (128, 4, 32), (32, 8, 32),
+                         (128, 1, 32), (256, 1, 32)]:
+        k = H.JitLapStage(
+            rk_o, tmp_o, red_o, sm1._ring_field_args, [], (h,) * 3,
+            grid, dx, nf, f_name=f_name,
+            name=f"tune_ls_{tile[0]}_{tile[1]}_{tile[2]}", tile=tile)
+        ms = timeit(lambda: k(env3))
+        bw = 16 * GBms / ms / nscalars
+        results.append((ms, tile))
+        print(f"  tile={tile}:  {ms:7.3f} ms   {bw:5.2f} TB/s")
+    results.sort()
+    print("  BEST:", results[0])
+
     # --- AOT gradlap XCHUNK sweep (env var)
     derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=grid)
     print(f"== AOT gradlap lap-only (4 passes ~ {4*GBms:.1f} GB)")
