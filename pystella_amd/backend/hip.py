@@ -787,26 +787,83 @@ extern "C" __global__ __launch_bounds__(TBZ * TBY) void {name}(
 """ + REDUCTION_TAIL
 
 
+class _NTLapCodegen(_LapCodegen):
+    """_LapCodegen plus read-redirection to nontemporal preloads."""
+
+    def __init__(self, *args, **kwargs):
+        super().__init__(*args, **kwargs)
+        self.preload = {}       # (name, lin) -> register var
+        self.store_ctx = False
+
+    def field_access(self, f, outer_idx):
+        if (not self.store_ctx and f.is_spatial
+                and len(outer_idx) <= 1 and not any(f.shift)):
+            lin = int(outer_idx[0]) if outer_idx else 0
+            reg = self.preload.get((f.name, lin))
+            if reg is not None:
+                return reg
+        return super().field_access(f, outer_idx)
+
+
 class JitLapStage:
     """Ring-buffer RK stage kernel fused with input-state reductions:
     the stencil field marches along x through a per-thread register ring
     (one new load per site instead of 4h+1), the Laplacian is formed in
     registers, the reducers accumulate input-state values, and the
-    2N-storage update stores last.  This is the fastest form of the hot
-    loop's only kernel (see fusion.StencilRKStepper)."""
+    2N-storage update stores last.  The k-arrays are read once via
+    nontemporal loads and every store is nontemporal (streaming data
+    never pollutes the L2, which stays available for stencil-neighbor
+    reuse).  This is the hot loop's only kernel
+    (see fusion.StencilRKStepper)."""
 
     def __init__(self, map_dict, tmp_instructions, entries, field_args,
                  scalar_names, halo, rank_shape, dx, nf, f_name="f",
-                 lap_name="lap_f", name="rk_lapstage", tile=(64, 4, 32)):
+                 lap_name="lap_f", name="rk_lapstage", tile=(256, 1, 32),
+                 nt=True):
         from pystella_amd.derivs import _LAP_COEFS
+        from pystella_amd.field import (
+            Field, Subscript, iter_exprs, walk_expr)
         self.rank_shape = tuple(rank_shape)
         self.tile = tile
         self.entries = entries
         h = max(halo) if isinstance(halo, (tuple, list)) else halo
         self.nf = nf
-        cg = _LapCodegen(field_args, halo, rank_shape, f_name, lap_name)
+        cg = _NTLapCodegen(field_args, halo, rank_shape, f_name, lap_name)
 
         lines = []
+        if nt:
+            # preload every read component of the unpadded streaming
+            # arrays (the RK k-arrays) with a nontemporal load
+            nt_names = {fa.name for fa in field_args
+                        if fa.spatial and not fa.padded
+                        and fa.name != lap_name
+                        and len(fa.outer_shape) <= 1}
+            used = set()
+
+            def visit(x):
+                if isinstance(x, Subscript) and \
+                        isinstance(x.aggregate, Field) and \
+                        x.aggregate.name in nt_names and \
+                        len(x.index) == 1 and \
+                        isinstance(x.index[0], int):
+                    used.add((x.aggregate.name, int(x.index[0])))
+                elif isinstance(x, Field) and x.name in nt_names \
+                        and not x.shape:
+                    used.add((x.name, 0))
+
+            for e in iter_exprs([list((tmp_instructions or {}).values()),
+                                 [e for e, _ in entries]]):
+                walk_expr(e, visit)
+            for nm, lin in sorted(used):
+                reg = f"pl_{nm}_{lin}"
+                off = "(((long)i*NY + j)*NZ + k)"
+                if lin:
+                    off = f"({lin}L*UVOL + {off})"
+                lines.append(
+                    f"const double {reg} = "
+                    f"__builtin_nontemporal_load(&{nm}[{off}]);")
+                cg.preload[(nm, lin)] = reg
+
         for lhs, rhs in (tmp_instructions or {}).items():
             tname = lhs.name if hasattr(lhs, "name") else str(lhs)
             cg.tmp_names.add(tname)
@@ -821,7 +878,15 @@ class JitLapStage:
                 "acc[%d] = %s; }" % (r, val, r, comb))
             combine_cases.append(f"(r == {r}) ? {comb} : ")
         for lhs, rhs in map_dict.items():
-            lines.append(f"{cg.emit(lhs)} = {cg.emit(rhs)};")
+            val = cg.emit(rhs)
+            cg.store_ctx = True
+            dst = cg.emit(lhs)
+            cg.store_ctx = False
+            if nt:
+                lines.append(
+                    f"__builtin_nontemporal_store({val}, &{dst});")
+            else:
+                lines.append(f"{dst} = {val};")
         combine = "".join(combine_cases) + "0.0"
 
         inv2 = [1.0 / d / d for d in dx]
@@ -874,11 +939,11 @@ class JitLapStage:
 def get_lap_stage_kernel(map_dict, tmp_instructions, entries, field_args,
                          scalar_names, halo, rank_shape, dx, nf,
                          f_name="f", lap_name="lap_f",
-                         name="rk_lapstage", tile=(64, 4, 32)):
+                         name="rk_lapstage", tile=(256, 1, 32), nt=True):
     return JitLapStage(map_dict, tmp_instructions, entries, field_args,
                        scalar_names, halo, rank_shape, dx, nf,
                        f_name=f_name, lap_name=lap_name, name=name,
-                       tile=tile)
+                       tile=tile, nt=nt)
 
 
 def get_lap_reduction_kernel(entries, field_args, scalar_names, halo,

@@ -332,9 +332,15 @@ class StencilRKStepper:
                         rhs_name = var(f"rhs_{i}")
                         tmp[rhs_name] = rhs_expr
                         tmp_o[rhs_name] = rhs_dict_orig[key]
-                        rk[k_acc] = (self_inner._A[stage] * k_acc
-                                     + dtv * rhs_name)
-                        rk_o[k_acc] = rk[k_acc]
+                        # keep the updated k in a register: one load and
+                        # one store of the k array per site, and stores
+                        # are never read back (safe for nontemporal)
+                        k_new = var(f"knew_{i}")
+                        tmp[k_new] = (self_inner._A[stage] * k_acc
+                                      + dtv * rhs_name)
+                        tmp_o[k_new] = tmp[k_new]
+                        rk[k_acc] = k_new
+                        rk_o[k_acc] = k_new
                         if ff.name in _Fused.pingpong:
                             out_f = Field(f"{ff.name}_next",
                                           offset=ff.offset,
@@ -343,7 +349,7 @@ class StencilRKStepper:
                             out_acc = out_f[outer] if outer else out_f
                         else:
                             out_acc = key
-                        rk[out_acc] = key + self_inner._B[stage] * k_acc
+                        rk[out_acc] = key + self_inner._B[stage] * k_new
                         rk_o[out_acc] = rk[out_acc]
                     if reduction is not None:
                         ring = None

@@ -59,6 +59,14 @@ def main(n=512):
     a = np.ones(1)
     hub = np.zeros(1)
 
+    # raw streaming roofline: copy (1R+1W) via torch
+    src = torch.rand(512**3 * 2, dtype=torch.float64, device=device)
+    dst = torch.empty_like(src)
+    ms = timeit(lambda: dst.copy_(src), n=20)
+    gb = src.numel() * 8 * 2 / 1e9
+    print(f"== torch copy roofline: {ms:.3f} ms  {gb/ms:.2f} TB/s "
+          f"(read+write combined)")
+
     tiles = [(64, 4, 512), (64, 4, 128), (64, 4, 64), (64, 4, 32),
              (64, 8, 64), (64, 8, 32), (128, 2, 64), (64, 2, 64),
              (256, 1, 64), (64, 16, 32)]
@@ -144,16 +152,20 @@ def main(n=512):
     print(f"== ring energy-fused rk stage kernel (16 passes ~ "
           f"{16*GBms/nscalars:.1f} GB)")
     results = []
-    for tile in tiles + [(128, 2, 32), (128, 4, 32), (32, 8, 32),
-                         (128, 1, 32), (256, 1, 32)]:
-        k = H.JitLapStage(
-            rk_o, tmp_o, red_o, sm1._ring_field_args, [], (h,) * 3,
-            grid, dx, nf, f_name=f_name,
-            name=f"tune_ls_{tile[0]}_{tile[1]}_{tile[2]}", tile=tile)
-        ms = timeit(lambda: k(env3))
-        bw = 16 * GBms / ms / nscalars
-        results.append((ms, tile))
-        print(f"  tile={tile}:  {ms:7.3f} ms   {bw:5.2f} TB/s")
+    for nt in (True, False):
+        for tile in tiles + [(128, 2, 32), (128, 4, 32), (32, 8, 32),
+                             (128, 1, 32), (256, 1, 32), (256, 1, 16),
+                             (512, 1, 32), (256, 2, 32)]:
+            k = H.JitLapStage(
+                rk_o, tmp_o, red_o, sm1._ring_field_args, [], (h,) * 3,
+                grid, dx, nf, f_name=f_name, nt=nt,
+                name=f"tune_ls{int(nt)}_{tile[0]}_{tile[1]}_{tile[2]}",
+                tile=tile)
+            ms = timeit(lambda: k(env3))
+            bw = 16 * GBms / ms / nscalars
+            results.append((ms, nt, tile))
+            print(f"  nt={int(nt)} tile={tile}:  {ms:7.3f} ms   "
+                  f"{bw:5.2f} TB/s")
     results.sort()
     print("  BEST:", results[0])
 
