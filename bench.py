@@ -47,6 +47,10 @@ def main():
                         help="keep the energy reduction as a separate "
                              "fused lap+reduce kernel instead of folding "
                              "it into the RK stage kernel")
+    parser.add_argument("--no-device-friedmann", action="store_true",
+                        help="run the Friedmann (a, adot) update on the "
+                             "host (one sync per RK stage) instead of "
+                             "on-device")
     p = parser.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -134,8 +138,19 @@ def main():
     energy = compute_energy(1.)
     expand = ps.Expansion(energy["total"], ps.LowStorageRK54, mpl=mpl)
 
+    device_loop = None
+    if fuse_energy and on_gpu and not p.no_device_friedmann:
+        # fully device-resident step: stage kernel + partials finish +
+        # (RCCL allreduce) + on-device Friedmann ODE — zero host syncs
+        from pystella_amd.fusion import DeviceFriedmannLoop
+        device_loop = DeviceFriedmannLoop(
+            stepper, decomp, expand, grid_size, dt, mpl=mpl)
+
     def step():
         nonlocal energy
+        if device_loop is not None:
+            device_loop.step(arrays)
+            return
         for s in range(stepper.num_stages):
             if fuse_energy:
                 # the stage kernel itself returns the input-state

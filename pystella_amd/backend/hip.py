@@ -788,12 +788,21 @@ extern "C" __global__ __launch_bounds__(TBZ * TBY) void {name}(
 
 
 class _NTLapCodegen(_LapCodegen):
-    """_LapCodegen plus read-redirection to nontemporal preloads."""
+    """_LapCodegen plus read-redirection to nontemporal preloads and
+    device-state scalars (scalars read from a device buffer instead of
+    being passed by value — lets the whole RK step run without host
+    synchronization, see JitFriedmann)."""
 
-    def __init__(self, *args, **kwargs):
+    def __init__(self, *args, state_map=None, **kwargs):
         super().__init__(*args, **kwargs)
         self.preload = {}       # (name, lin) -> register var
         self.store_ctx = False
+        self.state_map = state_map or {}
+
+    def scalar_param(self, name, idx=()):
+        if name in self.state_map and not idx:
+            return f"state[{self.state_map[name]}]"
+        return super().scalar_param(name, idx)
 
     def field_access(self, f, outer_idx):
         if (not self.store_ctx and f.is_spatial
@@ -818,8 +827,8 @@ class JitLapStage:
 
     def __init__(self, map_dict, tmp_instructions, entries, field_args,
                  scalar_names, halo, rank_shape, dx, nf, f_name="f",
-                 lap_name="lap_f", name="rk_lapstage", tile=(256, 1, 32),
-                 nt=True):
+                 lap_name="lap_f", name="rk_lapstage", tile=(64, 8, 64),
+                 nt=True, state_map=None):
         from pystella_amd.derivs import _LAP_COEFS
         from pystella_amd.field import (
             Field, Subscript, iter_exprs, walk_expr)
@@ -828,7 +837,9 @@ class JitLapStage:
         self.entries = entries
         h = max(halo) if isinstance(halo, (tuple, list)) else halo
         self.nf = nf
-        cg = _NTLapCodegen(field_args, halo, rank_shape, f_name, lap_name)
+        self.state_map = state_map
+        cg = _NTLapCodegen(field_args, halo, rank_shape, f_name, lap_name,
+                           state_map=state_map)
 
         lines = []
         if nt:
@@ -901,7 +912,8 @@ class JitLapStage:
         lapc0 = coefs[0] * (inv2[0] + inv2[1] + inv2[2])
 
         # pointer params: stencil field first, then every other spatial
-        # field referenced by the statements or reducers
+        # field referenced by the statements or reducers, then the
+        # optional device-state scalar buffer
         self.ptr_names = [f_name] + sorted(
             fa.name for fa in field_args
             if fa.spatial and fa.name not in (f_name, lap_name))
@@ -910,6 +922,9 @@ class JitLapStage:
                            if n in by_name]
         ptr_params = ", ".join(
             f"double* __restrict__ {n}" for n in self.ptr_names)
+        if state_map:
+            self.ptr_names.append("state")
+            ptr_params += ", const double* __restrict__ state"
         dbl_params = ", ".join(f"double {c}" for c, _ in cg.scalars)
         params = ", ".join(x for x in (
             ptr_params, "double* __restrict__ partials", dbl_params) if x)
@@ -934,6 +949,120 @@ class JitLapStage:
 
     _finish = JitReduction._finish
     __call__ = JitLapReduction.__call__
+
+    def launch_only(self, env):
+        """Launch without finishing the reduction; returns the raw
+        per-block partials tensor [nred, nblk] (all stream-ordered, no
+        host synchronization)."""
+        dev = None
+        ptrs = []
+        for n in self.ptr_names:
+            t = _check_tensor(n, env[n])
+            dev = t.device
+            ptrs.append(t.data_ptr())
+        nred = len(self.entries)
+        if (self._partials is None
+                or self._partials.device != dev
+                or self._partials.shape[1] != self.nblk):
+            self._partials = torch.empty((nred, self.nblk),
+                                         dtype=torch.float64, device=dev)
+        doubles = [_resolve_scalar(env, k) for k in self.scalar_keys]
+        ext().jit_launch(self.key, self.grid[0], self.grid[1],
+                         self.grid[2], self.block, 1, 1, 0, _stream(),
+                         ptrs + [self._partials.data_ptr()], [], doubles)
+        return self._partials
+
+
+FRIEDMANN_TEMPLATE = """
+#define NRED {nred}
+#define NBLK {nblk}
+extern "C" __global__ __launch_bounds__(256) void {name}_sums(
+    const double* __restrict__ partials, double* __restrict__ sums)
+{{
+    __shared__ double sd[256];
+    const int r = blockIdx.x;
+    double acc = 0.0;
+    for (int c = threadIdx.x; c < NBLK; c += 256)
+        acc += partials[(long)r * NBLK + c];
+    sd[threadIdx.x] = acc;
+    __syncthreads();
+    for (int s = 128; s > 0; s >>= 1) {{
+        if ((int)threadIdx.x < s)
+            sd[threadIdx.x] += sd[threadIdx.x + s];
+        __syncthreads();
+    }}
+    if (threadIdx.x == 0) sums[r] = sd[0];
+}}
+
+extern "C" __global__ void {name}_step(
+    const double* __restrict__ sums, double* __restrict__ state,
+    double A, double B, double dt)
+{{
+    const double wt[NRED] = {{ {wt} }};
+    const double wp[NRED] = {{ {wp} }};
+    double E = 0.0, P = 0.0;
+    #pragma unroll
+    for (int r = 0; r < NRED; ++r) {{
+        const double avg = sums[r] * {inv_grid!r};
+        E += wt[r] * avg;
+        P += wp[r] * avg;
+    }}
+    const double a = state[0];
+    const double adot = state[1];
+    const double rhs_a = adot;
+    const double rhs_adot = {frw_coef!r} * a * a * (E - 3.0 * P) * a;
+    const double ka = A * state[2] + dt * rhs_a;
+    const double kd = A * state[3] + dt * rhs_adot;
+    const double an = a + B * ka;
+    const double adn = adot + B * kd;
+    state[0] = an;
+    state[1] = adn;
+    state[2] = ka;
+    state[3] = kd;
+    state[4] = adn / an;
+    state[5] = E;
+    state[6] = P;
+}}
+"""
+
+
+class JitFriedmann:
+    """Device-resident Friedmann update: finishes the stage kernel's
+    per-block energy partials into E and P and advances the 2N-storage
+    (a, adot) ODE — all on the stream, no host round trip.  The state
+    buffer layout is [a, adot, k_a, k_adot, hubble, energy, pressure]
+    (the stage kernels read a and hubble straight from it).
+
+    Host analogue: :class:`pystella_amd.Expansion` (which mirrors
+    reference pystella/expansion.py:28-176)."""
+
+    STATE_A, STATE_ADOT, STATE_HUBBLE = 0, 1, 4
+    STATE_ENERGY, STATE_PRESSURE = 5, 6
+
+    def __init__(self, nred, nblk, wt, wp, grid_size, mpl=1.,
+                 name="friedmann"):
+        import math as _math
+        self.nred = nred
+        self.nblk = nblk
+        frw_coef = 4 * _math.pi / 3 / mpl**2
+        src = FRIEDMANN_TEMPLATE.format(
+            nred=nred, nblk=nblk, name=name,
+            wt=", ".join(repr(float(w)) for w in wt),
+            wp=", ".join(repr(float(w)) for w in wp),
+            inv_grid=1.0 / grid_size, frw_coef=frw_coef)
+        self.source = src
+        self.key_sums = ext().jit_compile(src, name + "_sums")
+        self.key_step = ext().jit_compile(src, name + "_step")
+
+    def finish_sums(self, partials, sums):
+        ext().jit_launch(self.key_sums, self.nred, 1, 1, 256, 1, 1, 0,
+                         _stream(),
+                         [partials.data_ptr(), sums.data_ptr()], [], [])
+
+    def step(self, sums, state, A, B, dt):
+        ext().jit_launch(self.key_step, 1, 1, 1, 1, 1, 1, 0, _stream(),
+                         [sums.data_ptr(), state.data_ptr()], [],
+                         [float(A), float(B), float(dt)])
 
 
 def get_lap_stage_kernel(map_dict, tmp_instructions, entries, field_args,

@@ -25,7 +25,7 @@ from __future__ import annotations
 import numpy as np
 import torch
 
-__all__ = ["DFT", "BaseDFT", "fftfreq"]
+__all__ = ["DFT", "BaseDFT", "fftfreq", "get_sliced_momenta"]
 
 
 def fftfreq(n):
@@ -39,6 +39,24 @@ def fftfreq(n):
 
 def rfftfreq(n):
     return np.fft.rfftfreq(n, 1 / n)
+
+
+def get_sliced_momenta(grid_shape, dtype, slc, device="cpu"):
+    """Per-rank k-space momentum arrays for a rank owning the k-space
+    slices ``slc`` (reference pystella/fourier/dft.py:335-348).
+
+    Returns ``{"momenta_x": tensor, "momenta_y": ..., "momenta_z": ...}``
+    with the z axis using the halved r2c grid for real ``dtype``.
+    """
+    import torch
+    dtype = np.dtype(dtype)
+    k = [fftfreq(n) for n in grid_shape]
+    if dtype.kind == "f":
+        k[-1] = rfftfreq(grid_shape[-1])
+    names = ("momenta_x", "momenta_y", "momenta_z")
+    return {name: torch.as_tensor(np.ascontiguousarray(k_i[s_i]),
+                                  device=device)
+            for name, k_i, s_i in zip(names, k, slc)}
 
 
 class BaseDFT:

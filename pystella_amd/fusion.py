@@ -394,3 +394,134 @@ class StencilRKStepper:
     @property
     def tmp_arrays(self):
         return self._stepper.tmp_arrays
+
+
+class DeviceFriedmannLoop:
+    """Fully device-resident scalar-preheating RK step: per stage, the
+    energy-fused ring stage kernel (reading a/H from a device state
+    buffer), the partials finish, an optional RCCL all-reduce, and the
+    Friedmann (a, ȧ) ODE update all execute on the stream with ZERO host
+    synchronization.  The host only reads the state buffer when asked
+    (``read_state``).
+
+    Numerics are identical to the host loop (``Expansion.step`` +
+    stage-returned energies); a GPU test asserts this.
+
+    :arg stepper: an energy-fused :class:`StencilRKStepper`.
+    :arg expand: a host :class:`~pystella_amd.Expansion` providing the
+        initial (a, ȧ) state (its Stepper must be the same low-storage
+        tableau).
+    """
+
+    STATE_LEN = 8
+
+    def __init__(self, stepper, decomp, expand, grid_size, dt,
+                 mpl=1.0):
+        import copy
+
+        self.stepper = stepper
+        self.decomp = decomp
+        self.dt = dt
+        self.grid_size = float(grid_size)
+        self.mpl = mpl
+        red = stepper._reduction
+        if red is None:
+            raise ValueError("stepper must be built with reducers")
+        if any(op not in ("avg", "sum") for _, _, _, op in red.flat):
+            raise NotImplementedError(
+                "device Friedmann loop needs avg/sum reducers")
+        self._red = red
+        self._A = stepper._stepper._A
+        self._B = stepper._stepper._B
+        self.num_stages = stepper.num_stages
+
+        # linear weights of each reduction entry in (total, pressure):
+        # probe the callback with basis vectors (get_rho_and_p is linear)
+        nred = len(red.flat)
+        self.wt = np.zeros(nred)
+        self.wp = np.zeros(nred)
+        for r in range(nred):
+            vals = {key: np.zeros(len(entries))
+                    for key, entries in red.reducers.items()}
+            key_r, i_r, _, _ = red.flat[r]
+            vals[key_r][i_r] = 1.0
+            out = red.callback(copy.deepcopy(vals))
+            self.wt[r] = float(np.asarray(out["total"]).reshape(-1)[0])
+            self.wp[r] = float(
+                np.asarray(out["pressure"]).reshape(-1)[0])
+
+        # device state [a, adot, k_a, k_adot, hubble, energy, pressure]
+        self.state = None
+        self._expand0 = expand
+        self._fk = None
+        self._sums = None
+
+    def _ensure_state(self, device):
+        if self.state is None:
+            e = self._expand0
+            st = torch.zeros(self.STATE_LEN, dtype=torch.float64)
+            st[0] = float(e.a[0])
+            st[1] = float(e.adot[0])
+            st[4] = float(e.hubble[0])
+            self.state = st.to(device)
+            self._sums = torch.zeros(len(self._red.flat),
+                                     dtype=torch.float64, device=device)
+
+    def step(self, arrays, extra_scalars=None):
+        """One full RK step (num_stages stages); swaps the ping-pong
+        f/f_next entries of ``arrays`` in place."""
+        import torch.distributed as dist
+
+        f = arrays[next(iter(self.stepper.pingpong))]
+        self._ensure_state(f.device)
+        env = dict(arrays)
+        env["state"] = self.state
+        env["dt"] = self.dt
+        if extra_scalars:
+            env.update(extra_scalars)
+        for s in range(self.num_stages):
+            smap = self.stepper._stepper.steps[s]
+            if not self.stepper._stepper.tmp_arrays:
+                self.stepper._stepper.tmp_arrays = \
+                    self.stepper._stepper.get_tmp_arrays_like(**arrays)
+            env.update(self.stepper._stepper.tmp_arrays)
+            kern = self._stage_kernel(smap, env)
+            partials = kern.launch_only(env)
+            if self._fk is None:
+                from pystella_amd.backend.hip import JitFriedmann
+                self._fk = JitFriedmann(
+                    partials.shape[0], partials.shape[1], self.wt,
+                    self.wp, self.grid_size, mpl=self.mpl)
+            self._fk.finish_sums(partials, self._sums)
+            if self.decomp.nranks > 1:
+                dist.all_reduce(self._sums)
+            self._fk.step(self._sums, self.state, self._A[s],
+                          self._B[s], self.dt)
+            for name in self.stepper.pingpong:
+                arrays[name], arrays[f"{name}_next"] = \
+                    arrays[f"{name}_next"], arrays[name]
+                env[name] = arrays[name]
+                env[f"{name}_next"] = arrays[f"{name}_next"]
+                self.decomp.share_halos(arrays[name])
+
+    def _stage_kernel(self, smap, env):
+        m = smap._map
+        rank_shape = m._infer_rank_shape(env)
+        if smap._hip_kernel is None or \
+                smap._hip_kernel.rank_shape != rank_shape or \
+                getattr(smap._hip_kernel, "state_map", None) is None:
+            from pystella_amd.backend.hip import get_lap_stage_kernel
+            rk_o, tmp_o, red_o, f_name, nf = smap.ring
+            smap._hip_kernel = get_lap_stage_kernel(
+                rk_o, tmp_o, red_o, smap._ring_field_args, [],
+                m.halo_shape, rank_shape, smap.derivs.dx, nf,
+                f_name=f_name, lap_name=f"lap_{f_name}", name=m.name,
+                state_map={"a": 0, "hubble": 4})
+        return smap._hip_kernel
+
+    def read_state(self):
+        """Host-side snapshot {a, adot, hubble, energy, pressure} (one
+        sync)."""
+        st = self.state.cpu().numpy()
+        return {"a": st[0], "adot": st[1], "hubble": st[4],
+                "energy": st[5], "pressure": st[6]}
