@@ -28,10 +28,9 @@ _C_FUNCS = {
 
 
 def _c_double(x):
-    if isinstance(x, numbers.Integral):
-        return f"{float(x)!r}"
-    r = repr(float(x))
-    return r
+    # literal as a functional cast so that float-mode kernels
+    # (``using real = float``) do not silently promote to double
+    return f"real({float(x)!r})"
 
 
 class Codegen:
@@ -60,7 +59,7 @@ class Codegen:
         c_name = "s_" + c_name
         self._scalar_index[key] = c_name
         self.scalars.append((c_name, key))
-        return c_name
+        return f"real({c_name})"
 
     # ------------------------------------------------------------------
     def field_access(self, f: Field, outer_idx):
@@ -144,7 +143,7 @@ class Codegen:
         for lhs, rhs in (tmp_statements or {}).items():
             name = lhs.name if hasattr(lhs, "name") else str(lhs)
             self.tmp_names.add(name)
-            lines.append(f"const double {name} = {self.emit(rhs)};")
+            lines.append(f"const real {name} = {self.emit(rhs)};")
         for lhs, rhs in statements.items():
             lines.append(f"{self.emit(lhs)} = {self.emit(rhs)};")
         return "\n        ".join(lines)
@@ -152,25 +151,26 @@ class Codegen:
 
 PREAMBLE = """
 #define ps_pow1(x) (x)
-__device__ inline double ps_pow2(double x) { return x*x; }
-__device__ inline double ps_pow3(double x) { return x*x*x; }
-__device__ inline double ps_pow4(double x)
-{ double y = x*x; return y*y; }
-__device__ inline double ps_pow5(double x)
-{ double y = x*x; return y*y*x; }
-__device__ inline double ps_pow6(double x)
-{ double y = x*x; return y*y*y; }
-__device__ inline double ps_pow7(double x)
-{ double y = x*x; return y*y*y*x; }
-__device__ inline double ps_pow8(double x)
-{ double y = x*x; y = y*y; return y*y; }
+template <class T> __device__ inline T ps_pow2(T x) { return x*x; }
+template <class T> __device__ inline T ps_pow3(T x) { return x*x*x; }
+template <class T> __device__ inline T ps_pow4(T x)
+{ T y = x*x; return y*y; }
+template <class T> __device__ inline T ps_pow5(T x)
+{ T y = x*x; return y*y*x; }
+template <class T> __device__ inline T ps_pow6(T x)
+{ T y = x*x; return y*y*y; }
+template <class T> __device__ inline T ps_pow7(T x)
+{ T y = x*x; return y*y*y*x; }
+template <class T> __device__ inline T ps_pow8(T x)
+{ T y = x*x; y = y*y; return y*y; }
 """
 
 
-def geometry_defines(halo, rank_shape, max_h=None):
+def geometry_defines(halo, rank_shape, max_h=None, rtype="double"):
     h = max(halo) if isinstance(halo, (tuple, list)) else halo
     nx, ny, nz = rank_shape
     return f"""
+using real = {rtype};
 #define H {h}
 #define NX {nx}
 #define NY {ny}

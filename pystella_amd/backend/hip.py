@@ -110,22 +110,30 @@ def _tile_grid(tile, rank_shape):
             (nx + xchunk - 1) // xchunk)
 
 
+_RTYPE = {torch.float64: "double", torch.float32: "float"}
+
+
 class JitElementwise:
-    """Compiled fused per-site map over the interior grid."""
+    """Compiled fused per-site map over the interior grid.  Supports
+    fp64 and fp32 arrays (``using real = double|float`` baked into the
+    kernel; run-time scalars stay double and are cast on use)."""
 
     def __init__(self, map_dict, tmp_instructions, field_args, scalar_names,
-                 halo, rank_shape, name="ew_map", tile=(64, 4, 64)):
+                 halo, rank_shape, name="ew_map", tile=(64, 4, 64),
+                 dtype=torch.float64):
         self.rank_shape = tuple(rank_shape)
         self.tile = tile
+        self.dtype = dtype
         self.field_args = [fa for fa in field_args if fa.spatial]
         cg = Codegen(field_args, halo, rank_shape)
         body = cg.emit_statements(map_dict, tmp_instructions)
         ptr_params = ", ".join(
-            f"double* __restrict__ {fa.name}" for fa in self.field_args)
+            f"real* __restrict__ {fa.name}" for fa in self.field_args)
         dbl_params = ", ".join(f"double {c}" for c, _ in cg.scalars)
         params = ", ".join(x for x in (ptr_params, dbl_params) if x)
         src = ELEMENTWISE_TEMPLATE.format(
-            defines=geometry_defines(halo, rank_shape)
+            defines=geometry_defines(halo, rank_shape,
+                                     rtype=_RTYPE[dtype])
             + _tile_defines(tile, rank_shape),
             preamble=PREAMBLE, name=name, params=params, body=body)
         self.source = src
@@ -138,6 +146,10 @@ class JitElementwise:
         ptrs = []
         for fa in self.field_args:
             t = _check_tensor(fa.name, env[fa.name])
+            if t.dtype != self.dtype:
+                raise TypeError(
+                    f"argument {fa.name} has dtype {t.dtype}, kernel "
+                    f"compiled for {self.dtype}")
             ptrs.append(t.data_ptr())
         doubles = [_resolve_scalar(env, k) for k in self.scalar_keys]
         ext().jit_launch(self.key, self.grid[0], self.grid[1],
@@ -147,10 +159,10 @@ class JitElementwise:
 
 def get_elementwise_kernel(map_dict, tmp_instructions, field_args,
                            scalar_names, halo, rank_shape, name="ew_map",
-                           tile=(64, 4, 64)):
+                           tile=(64, 4, 64), dtype=torch.float64):
     return JitElementwise(map_dict, tmp_instructions, field_args,
                           scalar_names, halo, rank_shape, name=name,
-                          tile=tile)
+                          tile=tile, dtype=dtype)
 
 
 # ---------------------------------------------------------------------------
@@ -318,9 +330,12 @@ class JitReduction:
     finished with torch ops + one packed allreduce by the caller."""
 
     def __init__(self, entries, field_args, scalar_names, halo, rank_shape,
-                 name="reduce_map", tile=(64, 4, 64)):
+                 name="reduce_map", tile=(64, 4, 64), dtype=None):
+        import torch as _t
+        dtype = dtype if dtype is not None else _t.float64
         self.rank_shape = tuple(rank_shape)
         self.tile = tile
+        self.dtype = dtype
         self.entries = entries
         self.field_args = [fa for fa in field_args if fa.spatial]
         nred = len(entries)
@@ -340,13 +355,13 @@ class JitReduction:
         combine = "".join(combine_cases) + "0.0"
 
         ptr_params = ", ".join(
-            f"const double* __restrict__ {fa.name}"
+            f"const real* __restrict__ {fa.name}"
             for fa in self.field_args)
         dbl_params = ", ".join(f"double {c}" for c, _ in cg.scalars)
         params = ", ".join(x for x in (
             ptr_params, "double* __restrict__ partials", dbl_params) if x)
 
-        defines = geometry_defines(halo, rank_shape)
+        defines = geometry_defines(halo, rank_shape, rtype=_RTYPE[dtype])
         defines += _tile_defines(tile, rank_shape)
         defines += ("#define COMBINE(r, a, b) (" + combine + ")\n")
         src = REDUCTION_TEMPLATE.format(
@@ -402,9 +417,9 @@ JitStageReduction._finish = JitReduction._finish
 
 
 def get_reduction_kernel(entries, field_args, scalar_names, halo,
-                         rank_shape, tile=(64, 4, 64)):
+                         rank_shape, tile=(64, 4, 64), dtype=None):
     return JitReduction(entries, field_args, scalar_names, halo,
-                        rank_shape, tile=tile)
+                        rank_shape, tile=tile, dtype=dtype)
 
 
 # ---------------------------------------------------------------------------

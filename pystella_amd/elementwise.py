@@ -113,10 +113,19 @@ class ElementWiseMap:
     # ------------------------------------------------------------------
     def _call_hip(self, env, rank_shape):
         from pystella_amd.backend.hip import get_elementwise_kernel
+        dtype = None
+        for fa in self.field_args:
+            t = env.get(fa.name)
+            if isinstance(t, torch.Tensor) and fa.spatial:
+                dtype = t.dtype
+                break
+        if dtype is None:
+            dtype = torch.float64
         if self._hip_kernel is None or \
-                self._hip_kernel.rank_shape != rank_shape:
+                self._hip_kernel.rank_shape != rank_shape or \
+                self._hip_kernel.dtype != dtype:
             self._hip_kernel = get_elementwise_kernel(
                 self.map_dict, self.tmp_instructions, self.field_args,
                 sorted(self.scalar_names), self.halo_shape, rank_shape,
-                name=self.name)
+                name=self.name, dtype=dtype)
         self._hip_kernel(env)

@@ -133,12 +133,21 @@ class Reduction:
 
     def _local_hip(self, env, rank_shape):
         from pystella_amd.backend.hip import get_reduction_kernel
+        dtype = None
+        for fa in self.field_args:
+            t = env.get(fa.name)
+            if isinstance(t, torch.Tensor) and fa.spatial:
+                dtype = t.dtype
+                break
+        if dtype is None:
+            dtype = torch.float64
         if self._hip_kernel is None or \
-                self._hip_kernel.rank_shape != rank_shape:
+                self._hip_kernel.rank_shape != rank_shape or \
+                self._hip_kernel.dtype != dtype:
             self._hip_kernel = get_reduction_kernel(
                 [(expr, op) for _, _, expr, op in self.flat],
                 self.field_args, sorted(self.scalar_names),
-                self.halo_shape, rank_shape)
+                self.halo_shape, rank_shape, dtype=dtype)
         return self._hip_kernel(env)
 
     def __call__(self, queue=None, filter_args=False, **kwargs):
