@@ -439,3 +439,45 @@ def test_device_friedmann_loop_gpu(grid_shape=(32, 32, 32)):
         (state["a"], a_host)
     cut = (slice(None),) + (slice(h, -h),) * 3
     assert (f_dev[cut] - f_host[cut]).abs().max().item() < 1e-13
+
+
+@requires_gpu
+@pytest.mark.parametrize("tdtype", [torch.float64, torch.float32])
+def test_relax_gpu_matches_cpu(tdtype, n=32, h=1):
+    """Jacobi relaxation (multigrid smoother) on GPU vs CPU, fp64 and
+    fp32 (`using real = float` kernels)."""
+    from pystella_amd.multigrid import JacobiIterator
+    from pystella_amd.field import Field, shift_fields
+
+    decomp = ps.DomainDecomposition((1, 1, 1), h,
+                                    rank_shape=(n, n, n))
+    dx = (2 * np.pi / n,) * 3
+    f = Field("f", offset="h")
+    rho = Field("rho", offset="h")
+    lap = sum(
+        (shift_fields(f, tuple(s * int(mu == d) for mu in range(3)))
+         - 2 * f
+         + shift_fields(f, tuple(-s * int(mu == d) for mu in range(3))))
+        for d in range(3) for s in [1]) / var("dx")[0]**2
+    problems = {f: (lap, rho)}
+    solver = JacobiIterator(decomp, problems, halo_shape=h,
+                            fixed_parameters=dict(omega=0.8))
+
+    torch.manual_seed(7)
+    pad = (n + 2 * h,) * 3
+    rho_t = (torch.rand(pad, dtype=torch.float64) - 0.5).to(tdtype)
+    decomp.share_halos(rho_t)
+
+    def run(device):
+        ff = torch.zeros(pad, dtype=tdtype, device=device)
+        tmp = torch.zeros_like(ff)
+        solver(decomp, iterations=20, f=ff, tmp_f=tmp,
+               rho=rho_t.to(device), dx=np.array(dx))
+        return ff.cpu()
+
+    fc = run("cpu")
+    fg = run("cuda")
+    torch.cuda.synchronize()
+    tol = 1e-12 if tdtype == torch.float64 else 1e-4
+    denom = fc.abs().max().item() + 1e-30
+    assert (fg - fc).abs().max().item() / denom < tol

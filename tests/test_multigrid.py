@@ -147,3 +147,28 @@ def test_multigrid_solve(MG):
     want = want - want.mean()
     rel = (got - want).abs().max().item() / want.abs().max().item()
     assert rel < 0.05, rel
+
+
+def test_multigrid_solve_fp32():
+    """The whole MG stack also runs in fp32 (the 1024^3-scale
+    production configuration; kernels compile with `using real=float`
+    on the GPU and the torch path follows the array dtype)."""
+    n, h = 32, 1
+    decomp, dx, problems, f_exact, rho = _poisson_setup(n, h)
+    solver = NewtonIterator(decomp, problems, halo_shape=h,
+                            fixed_parameters=dict(omega=0.8))
+    mg = FullApproximationScheme(solver, halo_shape=h)
+    f = torch.zeros_like(f_exact, dtype=torch.float32)
+    rho32 = rho.to(torch.float32)
+    errs = mg(decomp, dx0=dx, cycle=v_cycle(10, 20, 2),
+              f=f, rho=rho32)
+    final = [e for lvl, e in errs if lvl == 0][-1]["f"]
+    initial = [e for lvl, e in errs if lvl == 0][0]["f"]
+    assert f.dtype == torch.float32
+    assert final[1] < 0.05 * initial[1], (initial, final)
+    got = f[h:-h, h:-h, h:-h].double()
+    want = f_exact[h:-h, h:-h, h:-h]
+    got = got - got.mean()
+    want = want - want.mean()
+    rel = (got - want).abs().max().item() / want.abs().max().item()
+    assert rel < 0.05, rel
