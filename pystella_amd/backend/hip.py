@@ -1083,11 +1083,12 @@ class JitFriedmann:
 def get_lap_stage_kernel(map_dict, tmp_instructions, entries, field_args,
                          scalar_names, halo, rank_shape, dx, nf,
                          f_name="f", lap_name="lap_f",
-                         name="rk_lapstage", tile=(256, 1, 32), nt=True):
+                         name="rk_lapstage", tile=(64, 8, 64), nt=True,
+                         state_map=None):
     return JitLapStage(map_dict, tmp_instructions, entries, field_args,
                        scalar_names, halo, rank_shape, dx, nf,
                        f_name=f_name, lap_name=lap_name, name=name,
-                       tile=tile, nt=nt)
+                       tile=tile, nt=nt, state_map=state_map)
 
 
 def get_lap_reduction_kernel(entries, field_args, scalar_names, halo,
