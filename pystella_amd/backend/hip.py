@@ -755,6 +755,28 @@ class JitLapReduction:
         return self._finish(dev)
 
 
+# Reduction tail with runtime partials geometry (sub-box launches of
+# the same kernel write disjoint slices of one shared partials buffer).
+REDUCTION_TAIL_BOX = """
+    __shared__ double sd[TBZ * TBY];
+    const int bid = bid0 + (blockIdx.z * gridDim.y + blockIdx.y)
+                    * gridDim.x + blockIdx.x;
+    for (int r = 0; r < NRED; ++r) {{
+        sd[threadIdx.x] = acc[r];
+        __syncthreads();
+        for (int s = (TBZ * TBY) / 2; s > 0; s >>= 1) {{
+            if ((int)threadIdx.x < s)
+                sd[threadIdx.x] = COMBINE(r, sd[threadIdx.x],
+                                          sd[threadIdx.x + s]);
+            __syncthreads();
+        }}
+        if (threadIdx.x == 0)
+            partials[(long)r * nblkT + bid] = sd[0];
+        __syncthreads();
+    }}
+}}
+"""
+
 LAPSTAGE_TEMPLATE = """{defines}
 {preamble}
 #define NRED {nred}
@@ -764,11 +786,11 @@ extern "C" __global__ __launch_bounds__(TBZ * TBY) void {name}(
 {{
     double acc[NRED];
     {init}
-    const int k = blockIdx.x * TBZ + (threadIdx.x % TBZ);
-    const int j = blockIdx.y * TBY + (threadIdx.x / TBZ);
-    const int i0 = blockIdx.z * XCHUNK;
-    const int i1 = (i0 + XCHUNK < NX) ? i0 + XCHUNK : NX;
-    if (k < NZ && j < NY) {{
+    const int k = k0b + blockIdx.x * TBZ + (threadIdx.x % TBZ);
+    const int j = j0b + blockIdx.y * TBY + (threadIdx.x / TBZ);
+    const int i0 = i0b + blockIdx.z * XCHUNK;
+    const int i1 = (i0 + XCHUNK < i1b) ? i0 + XCHUNK : i1b;
+    if (k < k1b && j < j1b) {{
         const long sx = PSY * PSZ;
         double ring[NF][2 * H + 1];
         #pragma unroll
@@ -799,7 +821,7 @@ extern "C" __global__ __launch_bounds__(TBZ * TBY) void {name}(
                     ring[fld][p] = ring[fld][p + 1];
         }}
     }}
-""" + REDUCTION_TAIL
+""" + REDUCTION_TAIL_BOX
 
 
 class _NTLapCodegen(_LapCodegen):
@@ -815,8 +837,10 @@ class _NTLapCodegen(_LapCodegen):
         self.state_map = state_map or {}
 
     def scalar_param(self, name, idx=()):
+        # state scalars are hoisted into per-thread registers once at
+        # kernel entry (st_<name>, see JitLapStage state prologue)
         if name in self.state_map and not idx:
-            return f"state[{self.state_map[name]}]"
+            return f"st_{name}"
         return super().scalar_param(name, idx)
 
     def field_access(self, f, outer_idx):
@@ -895,6 +919,8 @@ class JitLapStage:
             cg.tmp_names.add(tname)
             lines.append(f"const double {tname} = {cg.emit(rhs)};")
         init_lines, combine_cases = [], []
+        for nm, sidx in sorted((state_map or {}).items()):
+            init_lines.append(f"const double st_{nm} = state[{sidx}];")
         for r, (expr, op) in enumerate(entries):
             init_lines.append(f"acc[{r}] = {_OP_INIT[op]};")
             comb = _OP_COMBINE[op]
@@ -940,9 +966,13 @@ class JitLapStage:
         if state_map:
             self.ptr_names.append("state")
             ptr_params += ", const double* __restrict__ state"
+        int_params = ("const int k0b, const int k1b, const int j0b, "
+                      "const int j1b, const int i0b, const int i1b, "
+                      "const int nblkT, const int bid0")
         dbl_params = ", ".join(f"double {c}" for c, _ in cg.scalars)
         params = ", ".join(x for x in (
-            ptr_params, "double* __restrict__ partials", dbl_params) if x)
+            ptr_params, "double* __restrict__ partials", int_params,
+            dbl_params) if x)
 
         defines = geometry_defines(halo, rank_shape)
         defines += _tile_defines(tile, rank_shape)
@@ -963,29 +993,60 @@ class JitLapStage:
         self._partials = None
 
     _finish = JitReduction._finish
-    __call__ = JitLapReduction.__call__
 
-    def launch_only(self, env):
-        """Launch without finishing the reduction; returns the raw
-        per-block partials tensor [nred, nblk] (all stream-ordered, no
-        host synchronization)."""
-        dev = None
+    def _box_geometry(self, box):
+        """(grid, ints-prefix) for a sub-box launch; box =
+        (i0, i1, j0, j1, k0, k1) in interior coordinates."""
+        tbz, tby, xchunk = self.tile
+        i0, i1, j0, j1, k0, k1 = box
+        grid = ((k1 - k0 + tbz - 1) // tbz,
+                (j1 - j0 + tby - 1) // tby,
+                (i1 - i0 + xchunk - 1) // xchunk)
+        return grid, [k0, k1, j0, j1, i0, i1]
+
+    def box_nblk(self, box):
+        grid, _ = self._box_geometry(box)
+        return grid[0] * grid[1] * grid[2]
+
+    def launch_box(self, env, box, partials, bid0, nblk_tot):
+        """Launch over a sub-box, writing this launch's per-block
+        partials at column offset ``bid0`` of the shared ``partials``
+        buffer [nred, nblk_tot].  Stream-ordered, no host sync."""
         ptrs = []
         for n in self.ptr_names:
             t = _check_tensor(n, env[n])
-            dev = t.device
             ptrs.append(t.data_ptr())
+        grid, ints = self._box_geometry(box)
+        doubles = [_resolve_scalar(env, k) for k in self.scalar_keys]
+        ext().jit_launch(self.key, grid[0], grid[1], grid[2],
+                         self.block, 1, 1, 0, _stream(),
+                         ptrs + [partials.data_ptr()],
+                         ints + [nblk_tot, bid0], doubles)
+
+    def launch_only(self, env):
+        """Full-grid launch without finishing the reduction; returns the
+        raw per-block partials tensor [nred, nblk] (stream-ordered, no
+        host synchronization)."""
+        dev = None
+        for n in self.ptr_names:
+            dev = _check_tensor(n, env[n]).device
         nred = len(self.entries)
         if (self._partials is None
                 or self._partials.device != dev
                 or self._partials.shape[1] != self.nblk):
             self._partials = torch.empty((nred, self.nblk),
                                          dtype=torch.float64, device=dev)
-        doubles = [_resolve_scalar(env, k) for k in self.scalar_keys]
-        ext().jit_launch(self.key, self.grid[0], self.grid[1],
-                         self.grid[2], self.block, 1, 1, 0, _stream(),
-                         ptrs + [self._partials.data_ptr()], [], doubles)
+        nx, ny, nz = self.rank_shape
+        self.launch_box(env, (0, nx, 0, ny, 0, nz), self._partials,
+                        0, self.nblk)
         return self._partials
+
+    def __call__(self, env):
+        dev = None
+        for n in self.ptr_names:
+            dev = _check_tensor(n, env[n]).device
+        self.launch_only(env)
+        return self._finish(dev)
 
 
 FRIEDMANN_TEMPLATE = """

@@ -67,6 +67,24 @@ def get_size_start(N, size, rank):
     return n, start
 
 
+class _HaloHandle:
+    """Pending overlapped halo exchange (see share_halos_start)."""
+
+    def __init__(self, works, fills):
+        self.works = works
+        self.fills = fills
+        self._done = False
+
+    def finish(self):
+        if self._done:
+            return
+        for w in self.works:
+            w.wait()
+        for dst, src in self.fills:
+            dst.copy_(src)
+        self._done = True
+
+
 class DomainDecomposition:
     """Pencil/slab/3-D domain decomposition with halo exchange.
 
@@ -203,6 +221,59 @@ class DomainDecomposition:
                 hi_rank = self.rankID(self.rx + delta[0], self.ry + delta[1],
                                       self.rz + delta[2])
                 self._exchange_axis(fx, axis, h, (lo_rank, hi_rank))
+
+    def share_halos_start(self, fx):
+        """Overlap-friendly halo exchange for STAR stencils: wraps
+        single-rank axes in place immediately (stream-ordered) and posts
+        ALL remote-axis face exchanges in one batched non-blocking
+        group, so interior compute can run while xGMI transfers are in
+        flight.  Unlike :meth:`share_halos`, edge/corner halo values are
+        NOT propagated (the faces are sent concurrently) — valid only
+        for axis-aligned (star) stencil reads, e.g. the Laplacian hot
+        loop.  Returns a handle; call ``handle.finish()`` before any
+        kernel that reads the halos.
+        """
+        dist = _dist()
+        hx, hy, hz = self.halo_shape
+        px, py, pz = self.proc_shape
+        dim = fx.dim()
+        ops = []
+        fills = []
+        for ax_rel, (h, p) in enumerate(zip((hx, hy, hz), (px, py, pz))):
+            if h == 0:
+                continue
+            axis = dim - 3 + ax_rel
+            if p == 1:
+                self._wrap_axis(fx, axis, h)
+                continue
+            n = fx.shape[axis] - 2 * h
+
+            def face(lo, extent, axis=axis):
+                sl = [slice(None)] * dim
+                sl[axis] = slice(lo, lo + extent)
+                return fx[tuple(sl)]
+
+            send_lo = face(h, h).contiguous()
+            send_hi = face(n, h).contiguous()
+            recv_lo = torch.empty_like(send_lo)
+            recv_hi = torch.empty_like(send_hi)
+            delta = [0, 0, 0]
+            delta[ax_rel] = 1
+            lo_rank = self.rankID(self.rx - delta[0], self.ry - delta[1],
+                                  self.rz - delta[2])
+            hi_rank = self.rankID(self.rx + delta[0], self.ry + delta[1],
+                                  self.rz + delta[2])
+            ops += [
+                dist.P2POp(dist.irecv, recv_hi, hi_rank),
+                dist.P2POp(dist.isend, send_lo, lo_rank),
+                dist.P2POp(dist.irecv, recv_lo, lo_rank),
+                dist.P2POp(dist.isend, send_hi, hi_rank),
+            ]
+            fills.append((face(0, h), recv_lo))
+            fills.append((face(n + h, h), recv_hi))
+
+        works = dist.batch_isend_irecv(ops) if ops else []
+        return _HaloHandle(works, fills)
 
     # -- collectives --------------------------------------------------------
 

@@ -455,6 +455,10 @@ class DeviceFriedmannLoop:
         self._expand0 = expand
         self._fk = None
         self._sums = None
+        self._boxes = None
+        self._partials = None
+        self._nblks = None
+        self._nblk_tot = 0
 
     def _ensure_state(self, device):
         if self.state is None:
@@ -467,9 +471,43 @@ class DeviceFriedmannLoop:
             self._sums = torch.zeros(len(self._red.flat),
                                      dtype=torch.float64, device=device)
 
+    def _regions(self, rank_shape):
+        """Partition of the rank box into an interior (stencil-safe
+        without fresh halos along remote axes) plus up to 6 boundary
+        slabs.  Box format: (i0, i1, j0, j1, k0, k1)."""
+        nx, ny, nz = rank_shape
+        h = max(self.stepper._stepper.halo_shape) \
+            if not isinstance(self.stepper._stepper.halo_shape,
+                              int) else self.stepper._stepper.halo_shape
+        px, py, pz = self.decomp.proc_shape
+        rx = px > 1
+        ry = py > 1
+        rz = pz > 1
+        ix = (h if rx else 0, nx - h if rx else nx)
+        jy = (h if ry else 0, ny - h if ry else ny)
+        kz = (h if rz else 0, nz - h if rz else nz)
+        interior = (ix[0], ix[1], jy[0], jy[1], kz[0], kz[1])
+        slabs = []
+        if rx:
+            slabs.append((0, h, 0, ny, 0, nz))
+            slabs.append((nx - h, nx, 0, ny, 0, nz))
+        if ry:
+            slabs.append((ix[0], ix[1], 0, h, 0, nz))
+            slabs.append((ix[0], ix[1], ny - h, ny, 0, nz))
+        if rz:
+            slabs.append((ix[0], ix[1], jy[0], jy[1], 0, h))
+            slabs.append((ix[0], ix[1], jy[0], jy[1], nz - h, nz))
+        return interior, slabs
+
     def step(self, arrays, extra_scalars=None):
         """One full RK step (num_stages stages); swaps the ping-pong
-        f/f_next entries of ``arrays`` in place."""
+        f/f_next entries of ``arrays`` in place.
+
+        Per stage: post the (star-stencil) halo exchange of f, launch
+        the interior stage kernel so compute overlaps the xGMI
+        transfers, then the boundary slabs, then finish the energy
+        partials + RCCL all-reduce + on-device Friedmann update — all
+        stream-ordered with no host synchronization."""
         import torch.distributed as dist
 
         f = arrays[next(iter(self.stepper.pingpong))]
@@ -486,11 +524,34 @@ class DeviceFriedmannLoop:
                     self.stepper._stepper.get_tmp_arrays_like(**arrays)
             env.update(self.stepper._stepper.tmp_arrays)
             kern = self._stage_kernel(smap, env)
-            partials = kern.launch_only(env)
+
+            handles = [self.decomp.share_halos_start(arrays[name])
+                       for name in self.stepper.pingpong]
+            interior, slabs = self._regions(kern.rank_shape)
+            if self._partials is None or \
+                    self._boxes != (interior, tuple(slabs)):
+                self._boxes = (interior, tuple(slabs))
+                nblks = [kern.box_nblk(b) for b in (interior, *slabs)]
+                self._nblks = nblks
+                self._nblk_tot = sum(nblks)
+                self._partials = torch.empty(
+                    (len(self._red.flat), self._nblk_tot),
+                    dtype=torch.float64, device=f.device)
+            partials = self._partials
+
+            kern.launch_box(env, interior, partials, 0, self._nblk_tot)
+            for h in handles:
+                h.finish()
+            bid0 = self._nblks[0]
+            for slab, nb in zip(slabs, self._nblks[1:]):
+                kern.launch_box(env, slab, partials, bid0,
+                                self._nblk_tot)
+                bid0 += nb
+
             if self._fk is None:
                 from pystella_amd.backend.hip import JitFriedmann
                 self._fk = JitFriedmann(
-                    partials.shape[0], partials.shape[1], self.wt,
+                    partials.shape[0], self._nblk_tot, self.wt,
                     self.wp, self.grid_size, mpl=self.mpl)
             self._fk.finish_sums(partials, self._sums)
             if self.decomp.nranks > 1:
@@ -502,7 +563,6 @@ class DeviceFriedmannLoop:
                     arrays[f"{name}_next"], arrays[name]
                 env[name] = arrays[name]
                 env[f"{name}_next"] = arrays[f"{name}_next"]
-                self.decomp.share_halos(arrays[name])
 
     def _stage_kernel(self, smap, env):
         m = smap._map
