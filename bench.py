@@ -47,6 +47,10 @@ def main():
                         help="keep the energy reduction as a separate "
                              "fused lap+reduce kernel instead of folding "
                              "it into the RK stage kernel")
+    parser.add_argument("--gws", action="store_true",
+                        help="include the gravitational-wave tensor "
+                             "sector (6 h_ij components sourced by the "
+                             "scalar stress tensor)")
     parser.add_argument("--no-device-friedmann", action="store_true",
                         help="run the Friedmann (a, adot) update on the "
                              "host (one sync per RK stage) instead of "
@@ -85,19 +89,22 @@ def main():
         return (mphi**2 / 2 * phi**2 + gsq / 2 * phi**2 * chi**2) / mphi**2
 
     sector = ps.ScalarSector(nscalars, potential=potential)
+    sectors = [sector]
+    if p.gws:
+        sectors.append(ps.TensorPerturbationSector([sector]))
     derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=rank_shape)
     from pystella_amd.fusion import (
         FusedLaplacianReduction, StencilRKStepper)
     fuse_energy = not (p.no_fuse or p.no_fuse_energy)
     if p.no_fuse:
-        stepper = ps.LowStorageRK54([sector], halo_shape=h,
+        stepper = ps.LowStorageRK54(sectors, halo_shape=h,
                                     rank_shape=rank_shape, dt=dt)
     elif fuse_energy:
         # fully fused MI355X structure: one kernel per RK stage that
         # evaluates the Laplacian inline (ping-pong f), updates the
         # unknowns AND reduces the input-state energy — no separate
         # energy pass in the hot loop at all
-        stepper = StencilRKStepper(ps.LowStorageRK54, [sector], derivs,
+        stepper = StencilRKStepper(ps.LowStorageRK54, sectors, derivs,
                                    halo_shape=h, rank_shape=rank_shape,
                                    dt=dt, reducers=sector,
                                    grid_size=grid_size,
@@ -105,7 +112,7 @@ def main():
     else:
         # stage kernels evaluate the Laplacian inline (ping-pong f);
         # the energy reduction is a separate fused lap+reduce kernel
-        stepper = StencilRKStepper(ps.LowStorageRK54, [sector], derivs,
+        stepper = StencilRKStepper(ps.LowStorageRK54, sectors, derivs,
                                    halo_shape=h, rank_shape=rank_shape,
                                    dt=dt)
     reduce_energy = FusedLaplacianReduction(
@@ -120,26 +127,41 @@ def main():
                                        dtype=torch.float64,
                                        generator=gen)).to(device)
     arrays = {"f": f, "dfdt": dfdt}
+    if p.gws:
+        hij = torch.zeros((6,) + pad, dtype=torch.float64, device=device)
+        arrays["hij"] = hij
+        arrays["dhijdt"] = torch.zeros_like(hij)
+        # GW rhs also reads the scalar gradients (stress tensor source)
+        arrays["dfdx"] = torch.zeros((nscalars, 3) + tuple(rank_shape),
+                                     dtype=torch.float64, device=device)
     if p.no_fuse:
         arrays["lap_f"] = torch.zeros(
             (nscalars,) + tuple(rank_shape), dtype=torch.float64,
             device=device)
+        if p.gws:
+            arrays["lap_hij"] = torch.zeros(
+                (6,) + tuple(rank_shape), dtype=torch.float64,
+                device=device)
     else:
         arrays["f_next"] = torch.zeros_like(f)
+        if p.gws:
+            arrays["hij_next"] = torch.zeros_like(hij)
 
     energy = None
 
     def compute_energy(a):
         # fused: halo exchange + inline Laplacian stencil + energy
         # reduction (one kernel)
-        kw = {k: v for k, v in arrays.items() if k != "f_next"}
+        kw = {k: v for k, v in arrays.items()
+              if not k.endswith("_next")}
         return reduce_energy(a=np.array(a), **kw)
 
     energy = compute_energy(1.)
     expand = ps.Expansion(energy["total"], ps.LowStorageRK54, mpl=mpl)
 
     device_loop = None
-    if fuse_energy and on_gpu and not p.no_device_friedmann:
+    if fuse_energy and on_gpu and not p.no_device_friedmann \
+            and not p.gws:
         # fully device-resident step: stage kernel + partials finish +
         # (RCCL allreduce) + on-device Friedmann ODE — zero host syncs
         from pystella_amd.fusion import DeviceFriedmannLoop
@@ -156,17 +178,27 @@ def main():
                 # the stage kernel itself returns the input-state
                 # energy — identical values to the reference loop's
                 # standalone reduction after the previous stage
+                if p.gws:
+                    derivs(fx=arrays["f"], grd=arrays["dfdx"])
                 energy = stepper(s, a=expand.a, hubble=expand.hubble,
                                  **arrays)
-                arrays["f"], arrays["f_next"] = \
-                    arrays["f_next"], arrays["f"]
-                decomp.share_halos(arrays["f"])
+                for name in stepper.pingpong:
+                    arrays[name], arrays[f"{name}_next"] = \
+                        arrays[f"{name}_next"], arrays[name]
+                    decomp.share_halos(arrays[name])
                 expand.step(s, energy["total"], energy["pressure"], dt)
             else:
+                if p.gws:
+                    derivs(fx=arrays["f"], grd=arrays["dfdx"])
+                    if p.no_fuse:
+                        derivs(fx=arrays["hij"],
+                               lap=arrays["lap_hij"])
                 stepper(s, a=expand.a, hubble=expand.hubble, **arrays)
                 if not p.no_fuse:
-                    arrays["f"], arrays["f_next"] = \
-                        arrays["f_next"], arrays["f"]
+                    for name in stepper.pingpong:
+                        arrays[name], arrays[f"{name}_next"] = \
+                            arrays[f"{name}_next"], arrays[name]
+                        decomp.share_halos(arrays[name])
                 expand.step(s, energy["total"], energy["pressure"], dt)
                 energy = compute_energy(expand.a)
 
@@ -204,7 +236,8 @@ def main():
             "dtype": "fp64",
             "data": "synthetic random-init fields",
             "config": {
-                "model": "scalar_preheating",
+                "model": ("scalar_preheating+gw" if p.gws
+                          else "scalar_preheating"),
                 "grid_shape": list(grid_shape),
                 "halo": h,
                 "nscalars": nscalars,
