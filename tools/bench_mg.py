@@ -1,0 +1,98 @@
+"""Poisson geometric-multigrid benchmark: 1024^3 fp32 on MI355X
+(BASELINE.json config #5).
+
+Manufactured-solution Poisson problem, FAS V-cycles with a Newton
+smoother; reports seconds per V-cycle and residual-reduction per cycle.
+"""
+
+import argparse
+import sys
+import time
+
+import numpy as np
+import torch
+
+sys.path.insert(0, ".")
+import pystella_amd as ps  # noqa: E402
+from pystella_amd.field import Field, shift_fields, var  # noqa: E402
+from pystella_amd.multigrid import (  # noqa: E402
+    FullApproximationScheme, NewtonIterator, v_cycle)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--n", type=int, default=1024)
+    ap.add_argument("--dtype", default="float32")
+    ap.add_argument("--cycles", type=int, default=4)
+    ap.add_argument("--depth", type=int, default=5)
+    ap.add_argument("--device", default=None)
+    p = ap.parse_args()
+
+    dtype = getattr(torch, p.dtype)
+    device = (torch.device(p.device) if p.device
+              else torch.device("cuda" if torch.cuda.is_available()
+                                else "cpu"))
+    n, h = p.n, 1
+    grid = (n, n, n)
+    decomp = ps.DomainDecomposition((1, 1, 1), h, rank_shape=grid)
+    L = 2 * np.pi
+    dx = (L / n,) * 3
+
+    f = Field("f", offset="h")
+    rho = Field("rho", offset="h")
+    lap = sum(
+        (shift_fields(f, tuple(s * int(mu == d) for mu in range(3)))
+         - 2 * f
+         + shift_fields(f, tuple(-s * int(mu == d) for mu in range(3))))
+        for d in range(3) for s in [1]) / var("dx")[0]**2
+    problems = {f: (lap, rho)}
+    solver = NewtonIterator(decomp, problems, halo_shape=h,
+                            fixed_parameters=dict(omega=0.8))
+    mg = FullApproximationScheme(solver, halo_shape=h)
+
+    # manufactured solution: f* = sin(x)sin(y)sin(z), rho = -3 f*
+    ax = torch.arange(n, dtype=torch.float64) * dx[0]
+    s1 = torch.sin(ax)
+    f_exact = (s1[:, None, None] * s1[None, :, None]
+               * s1[None, None, :])
+    pad = (n + 2 * h,) * 3
+    rho_t = torch.zeros(pad, dtype=torch.float64)
+    rho_t[h:-h, h:-h, h:-h] = -3.0 * f_exact
+    rho_t = rho_t.to(dtype).to(device)
+    decomp.share_halos(rho_t)
+
+    ff = torch.zeros(pad, dtype=dtype, device=device)
+    cyc = v_cycle(10, 20, p.depth)
+
+    def one_cycle():
+        return mg(decomp, dx0=dx, cycle=cyc, f=ff, rho=rho_t)
+
+    errs = one_cycle()   # warmup (includes setup + JIT)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(p.cycles):
+        errs = one_cycle()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    dtime = (time.perf_counter() - t0) / p.cycles
+
+    final = [e for lvl, e in errs if lvl == 0][-1]["f"]
+    initial = [e for lvl, e in errs if lvl == 0][0]["f"]
+    got = ff[h:-h, h:-h, h:-h].double().cpu()
+    got -= got.mean()
+    want = f_exact - f_exact.mean()
+    rel = (got - want).abs().max().item() / want.abs().max().item()
+    import json
+    print(json.dumps({
+        "metric": "seconds per FAS V-cycle, Poisson",
+        "value": dtime, "unit": "s", "higher_is_better": False,
+        "grid": list(grid), "dtype": p.dtype, "depth": p.depth,
+        "resid_L2_start": float(initial[1]),
+        "resid_L2_end": float(final[1]),
+        "rel_err_vs_exact": rel,
+    }))
+
+
+if __name__ == "__main__":
+    main()
