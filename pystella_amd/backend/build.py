@@ -15,7 +15,7 @@ PKG_DIR = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 CSRC = os.path.join(PKG_DIR, "csrc")
 SO_PATH = os.path.join(PKG_DIR, "_C.so")
 
-SOURCES = ["module.cpp", "derivs.hip"]
+SOURCES = ["module.cpp", "derivs.hip", "tt_mfma.hip"]
 ARCH = os.environ.get("PYSTELLA_ROCM_ARCH",
                       os.environ.get("PYTORCH_ROCM_ARCH", "gfx950"))
 

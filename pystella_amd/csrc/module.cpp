@@ -224,10 +224,27 @@ static int device_count()
     return n;
 }
 
+extern "C" void launch_tt_project_mfma(
+    const double *hij, double *out, const double *kx, const double *ky,
+    const double *kz, int ny, int nz, long vol, hipStream_t stream);
+
+static void tt_project(uintptr_t hij, uintptr_t out, uintptr_t kx,
+                       uintptr_t ky, uintptr_t kz, int ny, int nz,
+                       long vol, uintptr_t stream)
+{
+    launch_tt_project_mfma(
+        (const double *)hij, (double *)out, (const double *)kx,
+        (const double *)ky, (const double *)kz, ny, nz, vol,
+        (hipStream_t)stream);
+    HIP_CHECK(hipGetLastError());
+}
+
 PYBIND11_MODULE(_C, m)
 {
     m.doc() = "pystella_amd native runtime (gfx950)";
     m.def("gradlap", &gradlap, "fused gradient/Laplacian stencil");
+    m.def("tt_project", &tt_project,
+          "transverse-traceless projection via f64 MFMA");
     m.def("pd", &pd, "single-axis first derivative");
     m.def("jit_compile", &jit_compile, "compile HIP source via hiprtc");
     m.def("jit_launch", &jit_launch, "launch a JIT kernel");

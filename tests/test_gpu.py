@@ -481,3 +481,44 @@ def test_relax_gpu_matches_cpu(tdtype, n=32, h=1):
     tol = 1e-12 if tdtype == torch.float64 else 1e-4
     denom = fc.abs().max().item() + 1e-30
     assert (fg - fc).abs().max().item() / denom < tol
+
+
+@requires_gpu
+def test_tt_projection_mfma_vs_oracle(grid_shape=(16, 16, 16)):
+    """Transverse-traceless projection on the f64 matrix cores
+    (csrc/tt_mfma.hip) vs the torch oracle: transversality, traceless-
+    ness and exact agreement."""
+    from pystella_amd.fourier import DFT
+    decomp = ps.DomainDecomposition((1, 1, 1), 0,
+                                    rank_shape=grid_shape)
+    fft = DFT(decomp, grid_shape=grid_shape, dtype=np.float64,
+              device="cuda")
+    dk = (2 * np.pi / 5,) * 3
+    dx = (5 / grid_shape[0],) * 3
+    proj = ps.Projector(fft, 1, dk, dx)
+
+    kshape = fft.shape(True)
+    torch.manual_seed(9)
+    hij = (torch.randn((6,) + kshape, dtype=torch.float64)
+           + 1j * torch.randn((6,) + kshape, dtype=torch.float64)
+           ).to(torch.complex128)
+
+    # oracle: torch path (runs when tensors are CPU)
+    want = hij.clone()
+    proj_cpu = ps.Projector(
+        DFT(decomp, grid_shape=grid_shape, dtype=np.float64,
+            device="cpu"), 1, dk, dx)
+    proj_cpu.transverse_traceless(want)
+
+    got = hij.clone().cuda().contiguous()
+    proj.transverse_traceless(got)
+    torch.cuda.synchronize()
+    got = got.cpu()
+
+    err = (got - want).abs().max().item()
+    scale = want.abs().max().item()
+    assert err < 1e-12 * max(scale, 1.0), (err, scale)
+
+    # tracelessness of the MFMA result
+    tr = got[0] + got[3] + got[5]
+    assert tr.abs().max().item() < 1e-11 * max(scale, 1.0)
