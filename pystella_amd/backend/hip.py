@@ -277,6 +277,95 @@ def get_stage_reduction_kernel(map_dict, tmp_instructions, entries,
 
 
 # ---------------------------------------------------------------------------
+# Fused periodic-wrap kernel: all single-rank axes' halo faces in ONE
+# launch (the torch slicing path costs ~6 kernel launches per field per
+# stage).  Axes are wrapped concurrently, so edge/corner halos are NOT
+# propagated — star-stencil contract, same as share_halos_start.
+
+WRAP_TEMPLATE = """{defines}
+extern "C" __global__ __launch_bounds__(256) void {name}(
+    real* __restrict__ f)
+{{
+    const long tid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    const long stride = (long)gridDim.x * blockDim.x;
+#if WRAPX
+    for (long t = tid; t < (long)NF * 2 * H * PSY * PSZ; t += stride) {{
+        const int c = (int)(t % PSZ);
+        long r = t / PSZ;
+        const int j = (int)(r % PSY);
+        r /= PSY;
+        const int p = (int)(r % (2 * H));
+        const long base = (r / (2 * H)) * PVOL;
+        const int xd = p < H ? p : NX + H + (p - H);
+        const int xs = p < H ? p + NX : p;
+        f[base + ((long)xd * PSY + j) * PSZ + c] =
+            f[base + ((long)xs * PSY + j) * PSZ + c];
+    }}
+#endif
+#if WRAPY
+    for (long t = tid; t < (long)NF * 2 * H * PSX * PSZ; t += stride) {{
+        const int c = (int)(t % PSZ);
+        long r = t / PSZ;
+        const int i = (int)(r % PSX);
+        r /= PSX;
+        const int p = (int)(r % (2 * H));
+        const long base = (r / (2 * H)) * PVOL;
+        const int yd = p < H ? p : NY + H + (p - H);
+        const int ys = p < H ? p + NY : p;
+        f[base + ((long)i * PSY + yd) * PSZ + c] =
+            f[base + ((long)i * PSY + ys) * PSZ + c];
+    }}
+#endif
+#if WRAPZ
+    for (long t = tid; t < (long)NF * 2 * H * PSX * PSY; t += stride) {{
+        long r = t;
+        const int j = (int)(r % PSY);
+        r /= PSY;
+        const int i = (int)(r % PSX);
+        r /= PSX;
+        const int p = (int)(r % (2 * H));
+        const long base = (r / (2 * H)) * PVOL;
+        const int zd = p < H ? p : NZ + H + (p - H);
+        const int zs = p < H ? p + NZ : p;
+        f[base + ((long)i * PSY + j) * PSZ + zd] =
+            f[base + ((long)i * PSY + j) * PSZ + zs];
+    }}
+#endif
+}}
+"""
+
+_wrap_cache = {}
+
+
+def wrap_star(fx, halo, wrap_axes):
+    """Periodic wrap of all ``wrap_axes`` halo faces of ``fx`` in one
+    kernel launch (star-stencil contract; see WRAP_TEMPLATE)."""
+    h = max(halo) if isinstance(halo, (tuple, list)) else halo
+    nxp, nyp, nzp = fx.shape[-3:]
+    rank_shape = (nxp - 2 * h, nyp - 2 * h, nzp - 2 * h)
+    nf = 1
+    for n in fx.shape[:-3]:
+        nf *= n
+    key = (rank_shape, h, nf, tuple(sorted(wrap_axes)), fx.dtype)
+    k = _wrap_cache.get(key)
+    if k is None:
+        defines = geometry_defines(h, rank_shape, rtype=_RTYPE[fx.dtype])
+        defines += f"#define NF {nf}\n"
+        for ax, nm in enumerate("XYZ"):
+            defines += f"#define WRAP{nm} {1 if ax in wrap_axes else 0}\n"
+        name = f"wrap_star_{h}_{nf}_" + "".join(
+            str(int(ax in wrap_axes)) for ax in range(3))
+        src = WRAP_TEMPLATE.format(defines=defines, name=name)
+        key_id = ext().jit_compile(src, name)
+        cells = max((nxp + 2 * h) * (nyp + 2 * h), 1) * 2 * h * nf
+        grid = min(4096, (cells + 255) // 256)
+        _wrap_cache[key] = k = (key_id, grid)
+    _check_tensor("f", fx)
+    ext().jit_launch(k[0], k[1], 1, 1, 256, 1, 1, 0, _stream(),
+                     [fx.data_ptr()], [], [])
+
+
+# ---------------------------------------------------------------------------
 # JIT'd simultaneous reductions
 
 REDUCTION_TAIL = """

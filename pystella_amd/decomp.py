@@ -239,12 +239,20 @@ class DomainDecomposition:
         dim = fx.dim()
         ops = []
         fills = []
+        wrap_axes = [ax for ax, (h, p) in enumerate(
+            zip((hx, hy, hz), (px, py, pz))) if h > 0 and p == 1]
+        wrapped_fused = False
+        if wrap_axes and isinstance(fx, torch.Tensor) and fx.is_cuda:
+            from pystella_amd.backend.hip import wrap_star
+            wrap_star(fx, self.halo_shape, wrap_axes)
+            wrapped_fused = True
         for ax_rel, (h, p) in enumerate(zip((hx, hy, hz), (px, py, pz))):
             if h == 0:
                 continue
             axis = dim - 3 + ax_rel
             if p == 1:
-                self._wrap_axis(fx, axis, h)
+                if not wrapped_fused:
+                    self._wrap_axis(fx, axis, h)
                 continue
             n = fx.shape[axis] - 2 * h
 

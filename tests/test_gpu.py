@@ -522,3 +522,36 @@ def test_tt_projection_mfma_vs_oracle(grid_shape=(16, 16, 16)):
     # tracelessness of the MFMA result
     tr = got[0] + got[3] + got[5]
     assert tr.abs().max().item() < 1e-11 * max(scale, 1.0)
+
+
+@requires_gpu
+def test_wrap_star_kernel(grid_shape=(12, 10, 14), h=2):
+    """Fused periodic-wrap kernel vs the torch slicing path (face
+    halos; star contract)."""
+    decomp = ps.DomainDecomposition((1, 1, 1), h, rank_shape=grid_shape)
+    pad = tuple(n + 2 * h for n in grid_shape)
+    torch.manual_seed(13)
+    base = torch.rand((3,) + pad, dtype=torch.float64)
+
+    ref = base.clone()
+    decomp.share_halos(ref)         # CPU slicing path (with corners)
+
+    got = base.clone().cuda()
+    handle = decomp.share_halos_start(got)   # GPU fused wrap kernel
+    handle.finish()
+    torch.cuda.synchronize()
+    got = got.cpu()
+
+    nx, ny, nz = grid_shape
+    regions = [
+        (slice(h, h + nx), slice(h, h + ny), slice(h, h + nz)),
+        (slice(0, h), slice(h, h + ny), slice(h, h + nz)),
+        (slice(h + nx, None), slice(h, h + ny), slice(h, h + nz)),
+        (slice(h, h + nx), slice(0, h), slice(h, h + nz)),
+        (slice(h, h + nx), slice(h + ny, None), slice(h, h + nz)),
+        (slice(h, h + nx), slice(h, h + ny), slice(0, h)),
+        (slice(h, h + nx), slice(h, h + ny), slice(h + nz, None)),
+    ]
+    for reg in regions:
+        r = (slice(None),) + reg
+        assert torch.equal(got[r], ref[r]), reg

@@ -1,0 +1,9 @@
+set -x
+cd "$GRAFT_REPO_ROOT"
+mkdir -p gpurun_out
+timeout 600 python -m pytest tests -m gpu -x -q > gpurun_out/pytest_gpu6.log 2>&1
+echo "pytest exit=$?"
+timeout 900 python tools/bench_mg.py --n 1024 --depth 5 --cycles 3 > gpurun_out/bench_mg.log 2>&1
+echo "mg exit=$?"
+timeout 300 python bench.py --steps 10 --warmup 3 > gpurun_out/bench6.log 2>&1
+echo "bench exit=$?"
