@@ -101,16 +101,15 @@ class _StageRedMap:
         self.reduction = reduction
         self.derivs = derivs
         self.lap_names = list(lap_names)
-        # ring = (rk_orig, tmp_orig, red_entries_orig, f_name, nf):
-        # the pre-substitution statements (lap accesses intact) for the
-        # register-ring GPU kernel; available when exactly one stencil
-        # field is involved
+        # ring = list of (rk_orig, tmp_orig, red_entries_orig, f_name,
+        # nf), one per stencil family: the pre-substitution statements
+        # (lap accesses intact) for the register-ring GPU kernels
         self.ring = ring
         if ring is not None:
             from pystella_amd.field import get_field_args
-            rk_o, tmp_o, red_o, _, _ = ring
-            self._ring_field_args = get_field_args(
-                [tmp_o, rk_o, [e for e, _ in red_o]])
+            self._ring_field_args = [
+                get_field_args([tmp_o, rk_o, [e for e, _ in red_o]])
+                for rk_o, tmp_o, red_o, _, _ in ring]
         # extend argument discovery with the reducer expressions
         from pystella_amd.field import (
             Field, Variable, collect_fields, get_field_args, iter_exprs,
@@ -133,6 +132,7 @@ class _StageRedMap:
             walk_expr(e, visit)
         self._map.scalar_names = scal
         self._hip_kernel = None
+        self._kernel_shape = None
 
     def __call__(self, queue=None, **kwargs):
         import torch as _torch
@@ -143,16 +143,21 @@ class _StageRedMap:
                      for v in env.values())
         if on_gpu:
             if self._hip_kernel is None or \
-                    self._hip_kernel.rank_shape != rank_shape:
+                    self._kernel_shape != rank_shape:
+                self._kernel_shape = rank_shape
                 if self.ring is not None:
                     from pystella_amd.backend.hip import (
                         get_lap_stage_kernel)
-                    rk_o, tmp_o, red_o, f_name, nf = self.ring
-                    self._hip_kernel = get_lap_stage_kernel(
-                        rk_o, tmp_o, red_o, self._ring_field_args, [],
-                        m.halo_shape, rank_shape, self.derivs.dx, nf,
-                        f_name=f_name, lap_name=f"lap_{f_name}",
-                        name=m.name)
+                    self._hip_kernel = [
+                        get_lap_stage_kernel(
+                            rk_o, tmp_o,
+                            red_o or [(0.0, "sum")],
+                            fargs, [], m.halo_shape, rank_shape,
+                            self.derivs.dx, nf, f_name=f_name,
+                            lap_name=f"lap_{f_name}",
+                            name=f"{m.name}_{f_name}")
+                        for (rk_o, tmp_o, red_o, f_name, nf), fargs
+                        in zip(self.ring, self._ring_field_args)]
                 else:
                     from pystella_amd.backend.hip import (
                         get_stage_reduction_kernel)
@@ -160,6 +165,14 @@ class _StageRedMap:
                         m.map_dict, m.tmp_instructions, self.red_entries,
                         m.field_args, sorted(m.scalar_names),
                         m.halo_shape, rank_shape, name=m.name)
+            if self.ring is not None:
+                local = None
+                for kern, (_, _, red_o, _, _) in zip(self._hip_kernel,
+                                                     self.ring):
+                    out = kern(env)
+                    if red_o:
+                        local = out
+                return local, rank_shape
             local = self._hip_kernel(env)
             return local, rank_shape
         # CPU oracle: reduction of the input state with a lap scratch,
@@ -291,18 +304,64 @@ class StencilRKStepper:
         lap_name_list = self._lap_names
         derivs_ref = derivs
         rhs_dict_orig = dict(rhs_dict)
-        # register-ring kernel eligibility: a single stencil field
-        ring_f_name = None
-        ring_nf = 0
-        if reduction is not None and len(lap_name_list) == 1:
-            ring_f_name = lap_name_list[0][len("lap_"):]
-            for fld in fields:
-                if fld.name == ring_f_name:
-                    ring_nf = fld.shape[0] if fld.shape else 1
-                    break
         red_entries_orig = ([(expr, op)
                              for _, _, expr, op in reduction.flat]
                             if reduction is not None else [])
+
+        # register-ring kernel grouping: partition the unknowns by
+        # stencil family (a DynamicField and its .dot companion), one
+        # ring kernel per family per stage; the energy reducers ride
+        # with the family whose lap they reference
+        ring_groups = None     # [(f_name, nf, dot_name)]
+        key_group = {}         # key field name -> group index
+        red_group = 0
+        if reduction is not None and lap_name_list:
+            by_name = {f.name: f for f in fields}
+            # the unknown itself may appear only as a KEY (e.g. the
+            # wave equation's h_ij has no bare-field RHS term)
+            for key in rhs_dict:
+                kf, _ = _field_of(key)
+                by_name.setdefault(kf.name, kf)
+            ring_groups = []
+            for lap_name in lap_name_list:
+                fname = lap_name[len("lap_"):]
+                F = by_name.get(fname)
+                if F is None or not isinstance(F, DynamicField):
+                    ring_groups = None
+                    break
+                nf = F.shape[0] if F.shape else 1
+                gi = len(ring_groups)
+                ring_groups.append((fname, nf))
+                key_group[fname] = gi
+                key_group[F.dot.name] = gi
+            if ring_groups is not None:
+                for key in rhs_dict:
+                    kf, _ = _field_of(key)
+                    if kf.name not in key_group:
+                        ring_groups = None    # unknown outside families
+                        break
+            if ring_groups is not None:
+                # each group's ORIGINAL rhs may reference only its own
+                # lap (cross-group lap reads would need the other ring)
+                for key, expr in rhs_dict_orig.items():
+                    kf, _ = _field_of(key)
+                    gi = key_group[kf.name]
+                    own_lap = f"lap_{ring_groups[gi][0]}"
+                    for fld in collect_fields([expr]):
+                        if fld.name.startswith("lap_") and \
+                                fld.name != own_lap:
+                            ring_groups = None
+                            break
+                    if ring_groups is None:
+                        break
+            if ring_groups is not None and red_entries_orig:
+                red_laps = {f.name for f in collect_fields(
+                    [e for e, _ in red_entries_orig])
+                    if f.name.startswith("lap_")}
+                for gi, (fname, _) in enumerate(ring_groups):
+                    if f"lap_{fname}" in red_laps:
+                        red_group = gi
+                        break
 
         class _Fused(Stepper):
             pingpong = set(self.pingpong)
@@ -321,26 +380,28 @@ class StencilRKStepper:
                 for stage in range(self_inner.num_stages):
                     tmp = dict(lap_tmps)
                     rk = {}
-                    tmp_o = {}
-                    rk_o = {}
+                    ngroups = len(ring_groups) if ring_groups else 1
+                    tmp_g = [{} for _ in range(ngroups)]
+                    rk_g = [{} for _ in range(ngroups)]
                     for i, (key, rhs_expr) in enumerate(
                             self_inner.rhs_dict.items()):
                         ff, outer = _field_of(key)
+                        gi = key_group.get(ff.name, 0)
                         k = Field(f"{ff.name}_tmp", offset=0,
                                   shape=ff.shape, indices=ff.indices)
                         k_acc = k[outer] if outer else k
                         rhs_name = var(f"rhs_{i}")
                         tmp[rhs_name] = rhs_expr
-                        tmp_o[rhs_name] = rhs_dict_orig[key]
+                        tmp_g[gi][rhs_name] = rhs_dict_orig[key]
                         # keep the updated k in a register: one load and
                         # one store of the k array per site, and stores
                         # are never read back (safe for nontemporal)
                         k_new = var(f"knew_{i}")
                         tmp[k_new] = (self_inner._A[stage] * k_acc
                                       + dtv * rhs_name)
-                        tmp_o[k_new] = tmp[k_new]
+                        tmp_g[gi][k_new] = tmp[k_new]
                         rk[k_acc] = k_new
-                        rk_o[k_acc] = k_new
+                        rk_g[gi][k_acc] = k_new
                         if ff.name in _Fused.pingpong:
                             out_f = Field(f"{ff.name}_next",
                                           offset=ff.offset,
@@ -350,12 +411,16 @@ class StencilRKStepper:
                         else:
                             out_acc = key
                         rk[out_acc] = key + self_inner._B[stage] * k_new
-                        rk_o[out_acc] = rk[out_acc]
+                        rk_g[gi][out_acc] = rk[out_acc]
                     if reduction is not None:
                         ring = None
-                        if ring_f_name is not None:
-                            ring = (rk_o, tmp_o, red_entries_orig,
-                                    ring_f_name, ring_nf)
+                        if ring_groups is not None:
+                            ring = [
+                                (rk_g[gi], tmp_g[gi],
+                                 red_entries_orig if gi == red_group
+                                 else [], fname, nf)
+                                for gi, (fname, nf)
+                                in enumerate(ring_groups)]
                         steps.append(_StageRedMap(
                             rk, tmp_instructions=tmp,
                             red_entries=red_entries,
@@ -565,19 +630,28 @@ class DeviceFriedmannLoop:
                 env[f"{name}_next"] = arrays[f"{name}_next"]
 
     def _stage_kernel(self, smap, env):
+        if smap.ring is None or len(smap.ring) != 1:
+            raise NotImplementedError(
+                "DeviceFriedmannLoop requires a single-family ring "
+                "stepper (use the host fused loop for multi-sector "
+                "runs)")
         m = smap._map
         rank_shape = m._infer_rank_shape(env)
-        if smap._hip_kernel is None or \
-                smap._hip_kernel.rank_shape != rank_shape or \
-                getattr(smap._hip_kernel, "state_map", None) is None:
+        kern = smap._hip_kernel
+        if isinstance(kern, list):
+            kern = None
+        if kern is None or kern.rank_shape != rank_shape or \
+                kern.state_map is None:
             from pystella_amd.backend.hip import get_lap_stage_kernel
-            rk_o, tmp_o, red_o, f_name, nf = smap.ring
-            smap._hip_kernel = get_lap_stage_kernel(
-                rk_o, tmp_o, red_o, smap._ring_field_args, [],
+            rk_o, tmp_o, red_o, f_name, nf = smap.ring[0]
+            kern = get_lap_stage_kernel(
+                rk_o, tmp_o, red_o, smap._ring_field_args[0], [],
                 m.halo_shape, rank_shape, smap.derivs.dx, nf,
                 f_name=f_name, lap_name=f"lap_{f_name}", name=m.name,
                 state_map={"a": 0, "hubble": 4})
-        return smap._hip_kernel
+            smap._hip_kernel = kern
+            smap._kernel_shape = rank_shape
+        return kern
 
     def read_state(self):
         """Host-side snapshot {a, adot, hubble, energy, pressure} (one

@@ -148,7 +148,7 @@ def main(n=512):
     # --- register-ring energy-fused stage kernel sweep (current
     # default hot-loop kernel)
     sm1 = fst._stepper.steps[1]
-    rk_o, tmp_o, red_o, f_name, nf = sm1.ring
+    rk_o, tmp_o, red_o, f_name, nf = sm1.ring[0]
     print(f"== ring energy-fused rk stage kernel (16 passes ~ "
           f"{16*GBms/nscalars:.1f} GB)")
     results = []
@@ -157,7 +157,7 @@ def main(n=512):
                              (128, 1, 32), (256, 1, 32), (256, 1, 16),
                              (512, 1, 32), (256, 2, 32)]:
             k = H.JitLapStage(
-                rk_o, tmp_o, red_o, sm1._ring_field_args, [], (h,) * 3,
+                rk_o, tmp_o, red_o, sm1._ring_field_args[0], [], (h,) * 3,
                 grid, dx, nf, f_name=f_name, nt=nt,
                 name=f"tune_ls{int(nt)}_{tile[0]}_{tile[1]}_{tile[2]}",
                 tile=tile)
