@@ -108,7 +108,8 @@ def main():
                                    halo_shape=h, rank_shape=rank_shape,
                                    dt=dt, reducers=sector,
                                    grid_size=grid_size,
-                                   callback=get_rho_and_p)
+                                   callback=get_rho_and_p,
+                                   inline_grad=p.gws)
     else:
         # stage kernels evaluate the Laplacian inline (ping-pong f);
         # the energy reduction is a separate fused lap+reduce kernel
@@ -131,9 +132,13 @@ def main():
         hij = torch.zeros((6,) + pad, dtype=torch.float64, device=device)
         arrays["hij"] = hij
         arrays["dhijdt"] = torch.zeros_like(hij)
-        # GW rhs also reads the scalar gradients (stress tensor source)
-        arrays["dfdx"] = torch.zeros((nscalars, 3) + tuple(rank_shape),
-                                     dtype=torch.float64, device=device)
+        if not fuse_energy:
+            # unfused paths read the scalar gradients (stress tensor
+            # source) from an array; the fused stage kernels compute
+            # them inline (inline_grad)
+            arrays["dfdx"] = torch.zeros(
+                (nscalars, 3) + tuple(rank_shape),
+                dtype=torch.float64, device=device)
     if p.no_fuse:
         arrays["lap_f"] = torch.zeros(
             (nscalars,) + tuple(rank_shape), dtype=torch.float64,
@@ -178,8 +183,7 @@ def main():
                 # the stage kernel itself returns the input-state
                 # energy — identical values to the reference loop's
                 # standalone reduction after the previous stage
-                if p.gws:
-                    derivs(fx=arrays["f"], grd=arrays["dfdx"])
+                # (gradients for the GW source are inlined)
                 energy = stepper(s, a=expand.a, hubble=expand.hubble,
                                  **arrays)
                 for name in stepper.pingpong:

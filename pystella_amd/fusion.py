@@ -223,7 +223,7 @@ class StencilRKStepper:
 
     def __init__(self, Stepper, input, derivs, halo_shape=0,
                  rank_shape=None, dt=None, reducers=None, grid_size=None,
-                 callback=None, **kwargs):
+                 callback=None, inline_grad=False, **kwargs):
         from pystella_amd.field import (
             DynamicField, Field, Subscript, substitute, collect_fields,
             var)
@@ -283,6 +283,45 @@ class StencilRKStepper:
                     lap_tmps[lv] = lap_expr
                     subs[lap_acc] = lv
         self.pingpong = sorted(set(self.pingpong))
+
+        # optionally inline spatial gradients of stencil unknowns
+        # (e.g. the GW stress-tensor source ∂_i f ∂_j f): the pd
+        # companion array never exists in HBM and the per-stage
+        # gradient pass disappears.  Same stencil coefficients as
+        # FiniteDifferencer (derivs.py _GRAD_COEFS).
+        grad_tmps = {}
+        subs_grad = {}
+        self.inline_grad = bool(inline_grad)
+        if inline_grad:
+            from pystella_amd.derivs import _GRAD_COEFS
+            gcoefs = _GRAD_COEFS[h]
+            referenced = {f.name for f in fields}
+            seen = set()
+            for key in rhs_dict:
+                kf, _ = _field_of(key)
+                if not isinstance(kf, DynamicField) or kf.name in seen:
+                    continue
+                seen.add(kf.name)
+                if kf.pd.name not in referenced:
+                    continue
+                for fld in range(kf.shape[0] if kf.shape else 1):
+                    access = kf[fld] if kf.shape else kf
+                    for mu in range(3):
+                        gexpr = centered_diff(
+                            access, gcoefs, direction=mu + 1,
+                            order=1) * (1.0 / dx[mu])
+                        gv = var(f"gradv_{kf.name}_{fld}_{mu}")
+                        grad_tmps[gv] = gexpr
+                        pd_acc = (kf.pd[fld, mu] if kf.shape
+                                  else kf.pd[mu])
+                        subs_grad[pd_acc] = gv
+            if subs_grad:
+                # the group ("orig") statements keep lap accesses but
+                # use the inlined gradients
+                rhs_dict = {k: substitute(v, subs_grad)
+                            for k, v in rhs_dict.items()}
+        self._grad_tmps = grad_tmps
+
         new_rhs = {k: substitute(v, subs) for k, v in rhs_dict.items()}
 
         # fused input-state reducers (energy each RK stage without a
@@ -315,6 +354,7 @@ class StencilRKStepper:
         ring_groups = None     # [(f_name, nf, dot_name)]
         key_group = {}         # key field name -> group index
         red_group = 0
+        grad_groups = set()    # group indices that use inline grads
         if reduction is not None and lap_name_list:
             by_name = {f.name: f for f in fields}
             # the unknown itself may appear only as a KEY (e.g. the
@@ -354,6 +394,25 @@ class StencilRKStepper:
                             break
                     if ring_groups is None:
                         break
+            if ring_groups is not None and grad_tmps:
+                # which groups' statements consume the inlined
+                # gradients (their kernels must define the tmps)
+                from pystella_amd.field import (
+                    Variable as _Var, walk_expr as _walk)
+                grad_groups.clear()
+                for key, expr in rhs_dict_orig.items():
+                    kf, _ = _field_of(key)
+                    gi = key_group[kf.name]
+                    hits = []
+
+                    def _v(x, hits=hits):
+                        if isinstance(x, _Var) and \
+                                x.name.startswith("gradv_"):
+                            hits.append(x.name)
+
+                    _walk(expr, _v)
+                    if hits:
+                        grad_groups.add(gi)
             if ring_groups is not None and red_entries_orig:
                 red_laps = {f.name for f in collect_fields(
                     [e for e, _ in red_entries_orig])
@@ -378,10 +437,11 @@ class StencilRKStepper:
                                         self_inner._unknowns}
                 steps = []
                 for stage in range(self_inner.num_stages):
-                    tmp = dict(lap_tmps)
+                    tmp = {**grad_tmps, **lap_tmps}
                     rk = {}
                     ngroups = len(ring_groups) if ring_groups else 1
-                    tmp_g = [{} for _ in range(ngroups)]
+                    tmp_g = [dict(grad_tmps) if gi in grad_groups
+                             else {} for gi in range(ngroups)]
                     rk_g = [{} for _ in range(ngroups)]
                     for i, (key, rhs_expr) in enumerate(
                             self_inner.rhs_dict.items()):
