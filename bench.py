@@ -75,7 +75,10 @@ def main():
     h = p.halo
     box = (5., 5., 5.)
     dx = tuple(L / N for L, N in zip(box, grid_shape))
-    dt = 0.1 * min(dx)
+    # 0.1 dx is the CFL-ish step at the flagship 512^3 (dt=9.8e-4);
+    # additionally cap by the Friedmann timescale (H ~ 30 in these
+    # units) so small sanity grids stay stable
+    dt = min(0.1 * min(dx), 1e-3)
     nscalars = 2
     mphi, mpl, gsq = 1.2e-6, 1.0, 2.5e-7
 

@@ -93,3 +93,33 @@ def test_scalar_preheating_golden(tmp_path):
     assert abs(constraint - 2.6929495300365147e-08) < 1e-3 * \
         2.6929495300365147e-08 + 1e-12, constraint
     assert abs(float(expand.a[0]) - 1.5573428265664833) < 1e-6
+
+
+def test_bench_distributed_cpu(tmp_path):
+    """The driver's exact multi-rank launch pattern against bench.py
+    (torchrun, 2 ranks, gloo on CPU, tiny grid)."""
+    import json
+    import subprocess
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [sys.executable, "-m", "torch.distributed.run",
+           "--standalone", "--local-addr", "127.0.0.1",
+           "--nnodes=1", "--nproc-per-node", "2",
+           "--redirects", "3", "--log-dir", str(tmp_path / "trlogs"),
+           os.path.join(repo, "bench.py"),
+           "--gpus", "2", "--steps", "2", "--warmup", "1",
+           "--grid", "16", "--device", "cpu"]
+    env = {k: v for k, v in os.environ.items()
+           if k not in ("RANK", "LOCAL_RANK", "WORLD_SIZE",
+                        "MASTER_ADDR", "MASTER_PORT", "LOCAL_WORLD_SIZE",
+                        "GROUP_RANK", "TORCHELASTIC_RUN_ID")}
+    out = subprocess.run(cmd, capture_output=True, text=True,
+                         timeout=600, cwd=repo, env=env)
+    import glob
+    logs = "\n".join(open(f).read() for f in glob.glob(
+        str(tmp_path / "trlogs" / "**" / "*.log"), recursive=True))
+    assert out.returncode == 0, (out.stderr[-1500:], logs[-1500:])
+    line = [ln for ln in logs.splitlines() if ln.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "decomp3d[2, 1, 1]"
+    assert d["value"] > 0

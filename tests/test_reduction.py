@@ -302,3 +302,32 @@ def test_stage_fused_gw_cpu(grid_shape=(8, 8, 8)):
     assert (arrays["dfdt"][cut] - du[cut]).abs().max().item() < 1e-14
     assert (arrays["hij"][cut] - hu[cut]).abs().max().item() < 1e-14
     assert (arrays["dhijdt"][cut] - hdu[cut]).abs().max().item() < 1e-14
+
+
+def test_device_loop_region_partition():
+    """The interior/boundary-slab partition used by the overlapped
+    device loop covers the rank box exactly once for every proc_shape
+    (pure geometry; the GPU path launches one kernel per region)."""
+    from pystella_amd.fusion import DeviceFriedmannLoop
+
+    class _D:
+        pass
+
+    for proc_shape in [(1, 1, 1), (2, 1, 1), (1, 2, 1), (1, 1, 2),
+                       (2, 2, 1), (2, 2, 2), (4, 2, 1)]:
+        loop = DeviceFriedmannLoop.__new__(DeviceFriedmannLoop)
+        loop.decomp = _D()
+        loop.decomp.proc_shape = proc_shape
+
+        class _S:
+            pass
+
+        loop.stepper = _S()
+        loop.stepper._stepper = _S()
+        loop.stepper._stepper.halo_shape = 2
+        rank_shape = (16, 12, 20)
+        interior, slabs = loop._regions(rank_shape)
+        count = np.zeros(rank_shape, dtype=int)
+        for (i0, i1, j0, j1, k0, k1) in [interior] + slabs:
+            count[i0:i1, j0:j1, k0:k1] += 1
+        assert (count == 1).all(), (proc_shape, count.min(), count.max())
