@@ -169,6 +169,41 @@ def main(n=512):
     results.sort()
     print("  BEST:", results[0])
 
+    # --- GW tensor-group ring kernel sweep
+    tensor = ps.TensorPerturbationSector([sector])
+    fstg = StencilRKStepper(ps.LowStorageRK54, [sector, tensor],
+                            derivs0, halo_shape=h, rank_shape=grid,
+                            dt=dt, reducers=sector, grid_size=sites,
+                            callback=get_rho_and_p, inline_grad=True)
+    smg = fstg._stepper.steps[1]
+    gi = [i for i, g in enumerate(smg.ring) if g[3] == "hij"][0]
+    rk_t, tmp_t, red_t, fname_t, nf_t = smg.ring[gi]
+    hij = torch.zeros((6,) + pad, dtype=torch.float64, device=device)
+    env4 = dict(a=a, hubble=hub, f=f, hij=hij,
+                hij_next=torch.zeros_like(hij),
+                dhijdt=torch.zeros_like(hij),
+                hij_tmp=torch.zeros((6,) + grid, dtype=torch.float64,
+                                    device=device),
+                dhijdt_tmp=torch.zeros((6,) + grid, dtype=torch.float64,
+                                       device=device),
+                dt=dt)
+    print("== GW tensor-group ring kernel (~50 comps "
+          f"~ {50*GBms/nscalars:.1f} GB)")
+    results = []
+    for tile in [(64, 8, 64), (64, 4, 32), (128, 2, 32), (256, 1, 32),
+                 (64, 8, 32), (128, 4, 32), (64, 16, 32), (64, 4, 16)]:
+        k = H.JitLapStage(
+            rk_t, tmp_t, red_t or [(0.0, "sum")],
+            smg._ring_field_args[gi], [], (h,) * 3, grid, dx, nf_t,
+            f_name=fname_t, lap_name=f"lap_{fname_t}",
+            name=f"tune_gw_{tile[0]}_{tile[1]}_{tile[2]}", tile=tile)
+        ms = timeit(lambda: k(env4), n=5)
+        results.append((ms, tile))
+        print(f"  tile={tile}:  {ms:7.3f} ms   "
+              f"{50*GBms/nscalars/ms:5.2f} TB/s")
+    results.sort()
+    print("  BEST:", results[0])
+
     # --- AOT gradlap XCHUNK sweep (env var)
     derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=grid)
     print(f"== AOT gradlap lap-only (4 passes ~ {4*GBms:.1f} GB)")
