@@ -953,13 +953,30 @@ class JitLapStage:
     reuse).  This is the hot loop's only kernel
     (see fusion.StencilRKStepper)."""
 
+    # tile candidates, best-first for large grids; smaller tiles win on
+    # small per-rank grids (multi-GPU strong scaling) where the big
+    # tile cannot fill 256 CUs with enough workgroups
+    TILE_CANDIDATES = ((64, 8, 64), (64, 4, 32), (64, 2, 16),
+                       (32, 2, 8))
+
+    @classmethod
+    def pick_tile(cls, rank_shape):
+        for tile in cls.TILE_CANDIDATES:
+            g = _tile_grid(tile, rank_shape)
+            if g[0] * g[1] * g[2] >= 2048:
+                return tile
+        return max(cls.TILE_CANDIDATES,
+                   key=lambda t: int(np.prod(_tile_grid(t, rank_shape))))
+
     def __init__(self, map_dict, tmp_instructions, entries, field_args,
                  scalar_names, halo, rank_shape, dx, nf, f_name="f",
-                 lap_name="lap_f", name="rk_lapstage", tile=(64, 8, 64),
+                 lap_name="lap_f", name="rk_lapstage", tile=None,
                  nt=True, state_map=None):
         from pystella_amd.derivs import _LAP_COEFS
         from pystella_amd.field import (
             Field, Subscript, iter_exprs, walk_expr)
+        if tile is None:
+            tile = self.pick_tile(rank_shape)
         self.rank_shape = tuple(rank_shape)
         self.tile = tile
         self.entries = entries
@@ -1233,7 +1250,7 @@ class JitFriedmann:
 def get_lap_stage_kernel(map_dict, tmp_instructions, entries, field_args,
                          scalar_names, halo, rank_shape, dx, nf,
                          f_name="f", lap_name="lap_f",
-                         name="rk_lapstage", tile=(64, 8, 64), nt=True,
+                         name="rk_lapstage", tile=None, nt=True,
                          state_map=None):
     return JitLapStage(map_dict, tmp_instructions, entries, field_args,
                        scalar_names, halo, rank_shape, dx, nf,
