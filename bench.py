@@ -93,7 +93,16 @@ def main():
 
     sector = ps.ScalarSector(nscalars, potential=potential)
     sectors = [sector]
-    if p.gws:
+    fuse_energy_pre = not (p.no_fuse or p.no_fuse_energy)
+    if p.gws and fuse_energy_pre:
+        # split the 6 independent h_ij components into two 3-component
+        # stencil families: halves each ring kernel's register
+        # footprint (the components are views of one parent array)
+        sectors.append(ps.TensorPerturbationSector(
+            [sector], components=(0, 1, 2), name="hij_a"))
+        sectors.append(ps.TensorPerturbationSector(
+            [sector], components=(3, 4, 5), name="hij_b"))
+    elif p.gws:
         sectors.append(ps.TensorPerturbationSector([sector]))
     derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=rank_shape)
     from pystella_amd.fusion import (
@@ -131,7 +140,17 @@ def main():
                                        dtype=torch.float64,
                                        generator=gen)).to(device)
     arrays = {"f": f, "dfdt": dfdt}
-    if p.gws:
+    if p.gws and fuse_energy:
+        hij = torch.zeros((6,) + pad, dtype=torch.float64, device=device)
+        hij_next = torch.zeros_like(hij)
+        dhijdt = torch.zeros_like(hij)
+        arrays["hij_a"] = hij[0:3]
+        arrays["hij_b"] = hij[3:6]
+        arrays["hij_a_next"] = hij_next[0:3]
+        arrays["hij_b_next"] = hij_next[3:6]
+        arrays["dhij_adt"] = dhijdt[0:3]
+        arrays["dhij_bdt"] = dhijdt[3:6]
+    elif p.gws:
         hij = torch.zeros((6,) + pad, dtype=torch.float64, device=device)
         arrays["hij"] = hij
         arrays["dhijdt"] = torch.zeros_like(hij)
@@ -152,7 +171,7 @@ def main():
                 device=device)
     else:
         arrays["f_next"] = torch.zeros_like(f)
-        if p.gws:
+        if p.gws and not fuse_energy:
             arrays["hij_next"] = torch.zeros_like(hij)
 
     energy = None

@@ -112,10 +112,19 @@ class TensorPerturbationSector(Sector):
     of the) stress tensor of the given sectors
     (reference sectors.py:170-208)."""
 
-    def __init__(self, sectors, **kwargs):
+    def __init__(self, sectors, components=None, **kwargs):
+        ncomp = 6 if components is None else len(components)
+        name = kwargs.pop("name", "hij")
         self.hij = kwargs.pop(
-            "hij", DynamicField("hij", offset="h", shape=(6,)))
+            "hij", DynamicField(name, offset="h", shape=(ncomp,)))
         self.sectors = sectors
+        # components: subset of the 6 symmetric-pair indices this
+        # sector's field carries (the h_ij components are mutually
+        # independent in the EOM, so the sector can be split into
+        # several smaller stencil families — on MI355X this halves the
+        # per-kernel register-ring footprint; see bench.py --gws)
+        self.components = (tuple(range(6)) if components is None
+                           else tuple(components))
 
     @property
     def rhs_dict(self):
@@ -124,7 +133,10 @@ class TensorPerturbationSector(Sector):
         rhs = {}
         for i in range(1, 4):
             for j in range(i, 4):
-                fld = tensor_index(i, j)
+                pair = tensor_index(i, j)
+                if pair not in self.components:
+                    continue
+                fld = self.components.index(pair)
                 Sij = sum(sector.stress_tensor(i, j, drop_trace=True)
                           for sector in self.sectors)
                 rhs[hij[fld]] = hij.dot[fld]
