@@ -51,6 +51,9 @@ def main():
                         help="include the gravitational-wave tensor "
                              "sector (6 h_ij components sourced by the "
                              "scalar stress tensor)")
+    parser.add_argument("--gws-no-split", action="store_true",
+                        help="keep the GW tensor sector as one "
+                             "6-component stencil family")
     parser.add_argument("--no-device-friedmann", action="store_true",
                         help="run the Friedmann (a, adot) update on the "
                              "host (one sync per RK stage) instead of "
@@ -94,7 +97,7 @@ def main():
     sector = ps.ScalarSector(nscalars, potential=potential)
     sectors = [sector]
     fuse_energy_pre = not (p.no_fuse or p.no_fuse_energy)
-    if p.gws and fuse_energy_pre:
+    if p.gws and fuse_energy_pre and not p.gws_no_split:
         # split the 6 independent h_ij components into two 3-component
         # stencil families: halves each ring kernel's register
         # footprint (the components are views of one parent array)
@@ -140,7 +143,7 @@ def main():
                                        dtype=torch.float64,
                                        generator=gen)).to(device)
     arrays = {"f": f, "dfdt": dfdt}
-    if p.gws and fuse_energy:
+    if p.gws and fuse_energy and not p.gws_no_split:
         hij = torch.zeros((6,) + pad, dtype=torch.float64, device=device)
         hij_next = torch.zeros_like(hij)
         dhijdt = torch.zeros_like(hij)
@@ -154,6 +157,8 @@ def main():
         hij = torch.zeros((6,) + pad, dtype=torch.float64, device=device)
         arrays["hij"] = hij
         arrays["dhijdt"] = torch.zeros_like(hij)
+        if fuse_energy:
+            arrays["hij_next"] = torch.zeros_like(hij)
         if not fuse_energy:
             # unfused paths read the scalar gradients (stress tensor
             # source) from an array; the fused stage kernels compute
