@@ -51,9 +51,12 @@ def main():
                         help="include the gravitational-wave tensor "
                              "sector (6 h_ij components sourced by the "
                              "scalar stress tensor)")
-    parser.add_argument("--gws-no-split", action="store_true",
-                        help="keep the GW tensor sector as one "
-                             "6-component stencil family")
+    parser.add_argument("--gws-split", action="store_true",
+                        help="split the GW tensor sector into two "
+                             "3-component stencil families (measured "
+                             "slower: the stress-source gradients are "
+                             "recomputed per family; kept as an A/B "
+                             "flag)")
     parser.add_argument("--no-device-friedmann", action="store_true",
                         help="run the Friedmann (a, adot) update on the "
                              "host (one sync per RK stage) instead of "
@@ -97,7 +100,7 @@ def main():
     sector = ps.ScalarSector(nscalars, potential=potential)
     sectors = [sector]
     fuse_energy_pre = not (p.no_fuse or p.no_fuse_energy)
-    if p.gws and fuse_energy_pre and not p.gws_no_split:
+    if p.gws and fuse_energy_pre and p.gws_split:
         # split the 6 independent h_ij components into two 3-component
         # stencil families: halves each ring kernel's register
         # footprint (the components are views of one parent array)
@@ -143,7 +146,7 @@ def main():
                                        dtype=torch.float64,
                                        generator=gen)).to(device)
     arrays = {"f": f, "dfdt": dfdt}
-    if p.gws and fuse_energy and not p.gws_no_split:
+    if p.gws and fuse_energy and p.gws_split:
         hij = torch.zeros((6,) + pad, dtype=torch.float64, device=device)
         hij_next = torch.zeros_like(hij)
         dhijdt = torch.zeros_like(hij)
