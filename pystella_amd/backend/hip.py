@@ -870,7 +870,7 @@ LAPSTAGE_TEMPLATE = """{defines}
 {preamble}
 #define NRED {nred}
 #define NF {nf}
-extern "C" __global__ __launch_bounds__(TBZ * TBY) void {name}(
+extern "C" __global__ __launch_bounds__(TBZ * TBY, MINW) void {name}(
     {params})
 {{
     double acc[NRED];
@@ -971,7 +971,7 @@ class JitLapStage:
     def __init__(self, map_dict, tmp_instructions, entries, field_args,
                  scalar_names, halo, rank_shape, dx, nf, f_name="f",
                  lap_name="lap_f", name="rk_lapstage", tile=None,
-                 nt=True, state_map=None):
+                 nt=True, state_map=None, min_waves=1):
         from pystella_amd.derivs import _LAP_COEFS
         from pystella_amd.field import (
             Field, Subscript, iter_exprs, walk_expr)
@@ -1082,6 +1082,7 @@ class JitLapStage:
 
         defines = geometry_defines(halo, rank_shape)
         defines += _tile_defines(tile, rank_shape)
+        defines += f"#define MINW {min_waves}\n"
         defines += f"#define COMBINE(r, a, b) ({combine})\n"
         defines += f"#define LAPC0 ({lapc0!r})\n"
         src = LAPSTAGE_TEMPLATE.format(

@@ -190,17 +190,19 @@ def main(n=512):
     print("== GW tensor-group ring kernel (~50 comps "
           f"~ {50*GBms/nscalars:.1f} GB)")
     results = []
-    for tile in [(64, 8, 64), (64, 4, 32), (128, 2, 32), (256, 1, 32),
-                 (64, 8, 32), (128, 4, 32), (64, 16, 32), (64, 4, 16)]:
-        k = H.JitLapStage(
-            rk_t, tmp_t, red_t or [(0.0, "sum")],
-            smg._ring_field_args[gi], [], (h,) * 3, grid, dx, nf_t,
-            f_name=fname_t, lap_name=f"lap_{fname_t}",
-            name=f"tune_gw_{tile[0]}_{tile[1]}_{tile[2]}", tile=tile)
-        ms = timeit(lambda: k(env4), n=5)
-        results.append((ms, tile))
-        print(f"  tile={tile}:  {ms:7.3f} ms   "
-              f"{50*GBms/nscalars/ms:5.2f} TB/s")
+    for minw in (1, 2, 3, 4):
+        for tile in [(64, 8, 64), (64, 8, 32), (128, 4, 32),
+                     (64, 4, 32)]:
+            k = H.JitLapStage(
+                rk_t, tmp_t, red_t or [(0.0, "sum")],
+                smg._ring_field_args[gi], [], (h,) * 3, grid, dx, nf_t,
+                f_name=fname_t, lap_name=f"lap_{fname_t}",
+                name=f"tune_gw{minw}_{tile[0]}_{tile[1]}_{tile[2]}",
+                tile=tile, min_waves=minw)
+            ms = timeit(lambda: k(env4), n=5)
+            results.append((ms, minw, tile))
+            print(f"  minw={minw} tile={tile}:  {ms:7.3f} ms   "
+                  f"{50*GBms/nscalars/ms:5.2f} TB/s")
     results.sort()
     print("  BEST:", results[0])
 
