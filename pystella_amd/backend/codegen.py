@@ -124,6 +124,11 @@ class Codegen:
                     return f"(1.0/ps_pow{-n}({base}))"
             return f"pow({base}, {self.emit(expr.exponent)})"
         if isinstance(expr, Call):
+            if expr.func == "parity":
+                # checkerboard color of the GLOBAL site: (i+j+k+off)&1,
+                # off = parity of the rank's global start offset
+                off = self.emit(expr.args[0]) if expr.args else "0"
+                return f"real(((i + j + k + (int)({off})) & 1))"
             fn = _C_FUNCS.get(expr.func)
             if fn is None:
                 raise NotImplementedError(f"function {expr.func}")

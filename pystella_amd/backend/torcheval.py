@@ -106,6 +106,18 @@ def eval_expr(expr, env, ctx):
         expo = eval_expr(expr.exponent, env, ctx)
         return base ** expo
     if isinstance(expr, Call):
+        if expr.func == "parity":
+            base = getattr(ctx, "_parity", None)
+            if base is None:
+                n = ctx.rank_shape
+                ii = torch.arange(n[0]).view(-1, 1, 1)
+                jj = torch.arange(n[1]).view(1, -1, 1)
+                kk = torch.arange(n[2]).view(1, 1, -1)
+                base = (ii + jj + kk).to(torch.float64)
+                ctx._parity = base
+            off = int(eval_expr(expr.args[0], env, ctx)) if expr.args \
+                else 0
+            return (base + off) % 2
         args = [eval_expr(a, env, ctx) for a in expr.args]
         if all(isinstance(a, numbers.Number) for a in args):
             import math

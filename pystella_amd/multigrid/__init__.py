@@ -11,6 +11,7 @@ from pystella_amd.multigrid.transfer import (  # noqa: F401
 )
 from pystella_amd.multigrid.relax import (  # noqa: F401
     RelaxationBase, JacobiIterator, NewtonIterator,
+    RedBlackIterator,
 )
 from pystella_amd.multigrid.solver import (  # noqa: F401
     FullApproximationScheme, MultiGridSolver, mu_cycle, v_cycle, w_cycle,
@@ -20,6 +21,7 @@ from pystella_amd.multigrid.solver import (  # noqa: F401
 __all__ = [
     "Injection", "FullWeighting", "LinearInterpolation",
     "CubicInterpolation", "JacobiIterator", "NewtonIterator",
+    "RedBlackIterator",
     "FullApproximationScheme", "MultiGridSolver",
     "v_cycle", "w_cycle", "f_cycle",
 ]

@@ -17,7 +17,8 @@ from pystella_amd.elementwise import ElementWiseMap
 from pystella_amd.field import Field, diff, fabs, var
 from pystella_amd.reduction import Reduction
 
-__all__ = ["RelaxationBase", "JacobiIterator", "NewtonIterator"]
+__all__ = ["RelaxationBase", "JacobiIterator", "NewtonIterator",
+           "RedBlackIterator"]
 
 
 class RelaxationBase:
@@ -139,3 +140,58 @@ class NewtonIterator(RelaxationBase):
         D = diff(lhs, f)
         omega = var("omega")
         return f - omega * (lhs - rho) / D
+
+
+class RedBlackIterator(RelaxationBase):
+    """Red-black Gauss–Seidel smoother: two masked in-place half-sweeps
+    per iteration, updating one checkerboard color from the freshly
+    updated other color.  Converges roughly twice as fast per sweep as
+    damped Jacobi and needs no ping-pong temporary.
+
+    Valid for odd-offset (h=1) star stencils only — same-color
+    neighbors would race the in-place update otherwise.  No analogue in
+    the reference (its smoothers are Jacobi/Newton, relax.py:323-373);
+    this is an MI355X-motivated addition (in-place halves the HBM
+    write+swap traffic of the smoothing loop).
+    """
+
+    def __init__(self, decomp, queue=None, lhs_dict=None, halo_shape=0,
+                 **kwargs):
+        super().__init__(decomp, queue=queue, lhs_dict=lhs_dict,
+                         halo_shape=halo_shape, **kwargs)
+        h3 = self._h3
+        if max(h3) != 1:
+            raise ValueError("RedBlackIterator requires halo_shape=1 "
+                             "(odd-offset star stencils)")
+        from pystella_amd.field import Call, Comparison, If
+        common = dict(halo_shape=halo_shape,
+                      rank_shape=getattr(self, "rank_shape", None),
+                      fixed_parameters=self.fixed_parameters)
+        off = var("rb_off")
+        self.color_steppers = []
+        for color in (0, 1):
+            step_dict = {}
+            for f, (lhs, rho) in self.lhs_dict.items():
+                gs = self.step_operator(f, lhs, rho)
+                cond = Comparison(Call("parity", (off,)), "==",
+                                  float(color))
+                step_dict[f] = If(cond, gs, f)
+            self.color_steppers.append(
+                ElementWiseMap(step_dict, name=f"rbgs_color{color}",
+                               **common))
+
+    def step_operator(self, f, lhs, rho):
+        D = diff(lhs, f)
+        omega = var("omega")
+        return f - omega * (lhs - rho) / D
+
+    def __call__(self, decomp, queue=None, iterations=100, **kwargs):
+        kwargs.pop("solve_constraint", None)
+        if "rb_off" not in kwargs:
+            kwargs["rb_off"] = 0.0
+        for _ in range(iterations):
+            for stepper in self.color_steppers:
+                stepper(**{k: v for k, v in kwargs.items()
+                           if not k.startswith("tmp_")})
+                for name in self.unknown_names:
+                    decomp.share_halos(kwargs[name])
