@@ -16,7 +16,7 @@ sys.path.insert(0, ".")
 import pystella_amd as ps  # noqa: E402
 from pystella_amd.field import Field, shift_fields, var  # noqa: E402
 from pystella_amd.multigrid import (  # noqa: E402
-    FullApproximationScheme, NewtonIterator, v_cycle)
+    FullApproximationScheme, NewtonIterator, RedBlackIterator, v_cycle)
 
 
 def main():
@@ -26,6 +26,8 @@ def main():
     ap.add_argument("--cycles", type=int, default=4)
     ap.add_argument("--depth", type=int, default=5)
     ap.add_argument("--device", default=None)
+    ap.add_argument("--smoother", default="rbgs",
+                    choices=("rbgs", "newton"))
     p = ap.parse_args()
 
     dtype = getattr(torch, p.dtype)
@@ -46,8 +48,12 @@ def main():
          + shift_fields(f, tuple(-s * int(mu == d) for mu in range(3))))
         for d in range(3) for s in [1]) / var("dx")[0]**2
     problems = {f: (lap, rho)}
-    solver = NewtonIterator(decomp, problems, halo_shape=h,
-                            fixed_parameters=dict(omega=0.8))
+    if p.smoother == "rbgs":
+        solver = RedBlackIterator(decomp, problems, halo_shape=h,
+                                  fixed_parameters=dict(omega=1.0))
+    else:
+        solver = NewtonIterator(decomp, problems, halo_shape=h,
+                                fixed_parameters=dict(omega=0.8))
     mg = FullApproximationScheme(solver, halo_shape=h)
 
     # manufactured solution: f* = sin(x)sin(y)sin(z), rho = -3 f*
@@ -88,6 +94,7 @@ def main():
         "metric": "seconds per FAS V-cycle, Poisson",
         "value": dtime, "unit": "s", "higher_is_better": False,
         "grid": list(grid), "dtype": p.dtype, "depth": p.depth,
+        "smoother": p.smoother,
         "resid_L2_start": float(initial[1]),
         "resid_L2_end": float(final[1]),
         "rel_err_vs_exact": rel,
