@@ -72,6 +72,21 @@ class RayleighGenerator:
         self.cdtype = fft.fk.dtype
 
     # ------------------------------------------------------------------
+    def _window_sq(self, window):
+        """window(k)² on the device (device-first, numpy fallback)."""
+        try:
+            w = window(self._kmags_t)
+            if isinstance(w, torch.Tensor) and \
+                    w.device == self._kmags_t.device:
+                return w.to(torch.float64) ** 2
+            if isinstance(w, (int, float)):
+                return float(w) ** 2
+            raise TypeError
+        except (TypeError, RuntimeError):
+            return torch.as_tensor(
+                np.asarray(window(self.kmags)) ** 2
+                * np.ones_like(self.kmags), device=self.device)
+
     def _uniform(self, n_sets):
         shape = (n_sets,) + self.kmags.shape
         u = torch.rand(shape, dtype=torch.float64, device=self.device,
@@ -94,13 +109,34 @@ class RayleighGenerator:
 
     def _ps_wrapper(self, ps_func, wk, kmags):
         """Evaluate a power spectrum, zeroing the homogeneous mode if
-        this rank holds it (reference rayleigh.py:174-185)."""
+        this rank holds it (reference rayleigh.py:174-185).
+
+        Arithmetic-only ``ps_func`` lambdas are evaluated directly on
+        the device tensor (no 0.5 GB host round trip per call at 512³);
+        functions that need numpy fall back to the host path."""
+        found_zero = kmags.flat[0] == 0. and np.all(
+            np.unravel_index(0, kmags.shape) == (0, 0, 0))
+        # device-first evaluation
+        try:
+            wk_t = (wk if isinstance(wk, torch.Tensor)
+                    else torch.as_tensor(wk, device=self.device))
+            wk_t = wk_t.clone()
+            if found_zero and kmags[0, 0, 0] == 0.:
+                wk_t[0, 0, 0] = wk_t[0, 0, 1]
+            power_t = ps_func(wk_t)
+            if not (isinstance(power_t, torch.Tensor)
+                    and power_t.device == wk_t.device):
+                raise TypeError
+            power_t = power_t.to(torch.float64) * torch.ones_like(wk_t)
+            if found_zero and kmags[0, 0, 0] == 0.:
+                power_t[0, 0, 0] = 0.
+            return power_t
+        except (TypeError, RuntimeError):
+            pass
         if isinstance(wk, torch.Tensor):
             wk_np = wk.cpu().numpy()
         else:
             wk_np = np.asarray(wk)
-        found_zero = kmags.flat[0] == 0. and np.all(
-            np.unravel_index(0, kmags.shape) == (0, 0, 0))
         wk_np = wk_np.copy()
         if found_zero and kmags[0, 0, 0] == 0.:
             wk_np[0, 0, 0] = wk_np[0, 0, 1]
@@ -122,10 +158,9 @@ class RayleighGenerator:
             rands[0] = np.exp(-1.)
 
         f_power = (amplitude_sq
-                   * torch.as_tensor(window(self.kmags) ** 2
-                                     * np.ones_like(self.kmags),
-                                     device=self.device)
-                   * self._ps_wrapper(field_ps, self.kmags, self.kmags))
+                   * self._window_sq(window)
+                   * self._ps_wrapper(field_ps, self._kmags_t,
+                                      self.kmags))
 
         amp = torch.sqrt(-torch.log(rands[0]))
         phs = torch.exp(2j * np.pi * rands[1])

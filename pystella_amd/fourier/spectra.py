@@ -53,16 +53,34 @@ class PowerSpectra:
         self._bin_idx = torch.as_tensor(
             np.round(kmags / self.bin_width).astype(np.int64),
             device=dev).reshape(-1).clamp_(0, self.num_bins - 1)
+        self._wbase_cache = {}
+        self._bidx32 = None
 
     def bin_power(self, fk, queue=None, k_power=3, allocator=None):
         """Unnormalized binned |f_k|² k^n with r2c double-count weights
-        (reference spectra.py:140-176)."""
-        w = (self._counts * self._kmags ** k_power
-             * torch.abs(fk) ** 2).reshape(-1)
-        hist = torch.zeros(self.num_bins, dtype=torch.float64,
-                           device=w.device)
-        hist.index_add_(0, self._bin_idx, w.to(torch.float64))
-        hist = hist.cpu().numpy()
+        (reference spectra.py:140-176).
+
+        GPU path: LDS-binned HIP kernel (torch index_add_ serializes on
+        the ~500 global bins and is ~500× slower at 512³)."""
+        if (isinstance(fk, torch.Tensor) and fk.is_cuda
+                and fk.dtype == torch.complex128 and fk.is_contiguous()
+                and self.num_bins <= 4096):
+            from pystella_amd.backend.hip import spectra_bin
+            wb = self._wbase_cache.get(k_power)
+            if wb is None:
+                wb = (self._counts * self._kmags ** k_power
+                      ).reshape(-1).to(torch.float64).contiguous()
+                self._wbase_cache[k_power] = wb
+                self._bidx32 = self._bin_idx.to(torch.int32).contiguous()
+            hist = spectra_bin(fk.reshape(-1), wb, self._bidx32,
+                               self.num_bins).cpu().numpy()
+        else:
+            w = (self._counts * self._kmags ** k_power
+                 * torch.abs(fk) ** 2).reshape(-1)
+            hist = torch.zeros(self.num_bins, dtype=torch.float64,
+                               device=w.device)
+            hist.index_add_(0, self._bin_idx, w.to(torch.float64))
+            hist = hist.cpu().numpy()
         hist = self.decomp.allreduce(hist)
         return hist / self.bin_counts
 
