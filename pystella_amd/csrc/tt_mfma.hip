@@ -61,9 +61,12 @@ __global__ __launch_bounds__(256) void tt_project_mfma(
     const int bj = lane & 3;          // B col
     double h_re = 0.0, h_im = 0.0;    // A = H for M1 = H.P
     if (ai < 3 && ak < 3) {
-        const double *p = hij + ((long)sym_idx(ai, ak) * vol + site) * 2;
-        h_re = p[0];
-        h_im = p[1];
+        typedef double d2 __attribute__((ext_vector_type(2)));
+        const d2 *p = (const d2 *)(hij
+            + ((long)sym_idx(ai, ak) * vol + site) * 2);
+        const d2 v = __builtin_nontemporal_load(p);
+        h_re = v.x;
+        h_im = v.y;
     }
     const double b_p = (ak < 3 && bj < 3) ? PEL(ak, bj) : 0.0;
     const double a_p = (ai < 3 && ak < 3) ? PEL(ai, ak) : 0.0;
@@ -95,9 +98,12 @@ __global__ __launch_bounds__(256) void tt_project_mfma(
         double re = m2_re - 0.5 * pab * tr_re;
         double im = m2_im - 0.5 * pab * tr_im;
         if (kzero) { re = 0.0; im = 0.0; }
-        double *o = out + ((long)sym_idx(ci, cj) * vol + site) * 2;
-        o[0] = re;
-        o[1] = im;
+        typedef double d2 __attribute__((ext_vector_type(2)));
+        d2 v;
+        v.x = re;
+        v.y = im;
+        __builtin_nontemporal_store(
+            v, (d2 *)(out + ((long)sym_idx(ci, cj) * vol + site) * 2));
     }
 #undef PEL
 }
