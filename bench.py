@@ -252,6 +252,14 @@ def main():
     # max over ranks
     elapsed = float(decomp.allreduce(elapsed, op="max"))
 
+    # a diverged (NaN) run must not be reported as a valid measurement
+    if device_loop is not None:
+        final_energy = device_loop.read_state()["energy"]
+    else:
+        final_energy = float(np.asarray(energy["total"]).reshape(-1)[0])
+    assert np.isfinite(final_energy), \
+        f"run diverged: energy={final_energy}"
+
     ms_per_step = elapsed / p.steps * 1e3
     msites = grid_size * p.steps / elapsed / 1e6
 
