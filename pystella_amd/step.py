@@ -54,6 +54,13 @@ def _prepend_index(expr, q):
             return Subscript(x, (q,))
         if isinstance(x, Subscript) and isinstance(x.aggregate, Field):
             return Subscript(x.aggregate, (q,) + x.index)
+        if isinstance(x, Subscript) and \
+                isinstance(x.aggregate, Subscript):
+            # map_expr rebuilds f[i] as Subscript(leaf(f), (i,)) =
+            # Subscript(Subscript(f, (q,)), (i,)); flatten to f[q, i]
+            inner = x.aggregate
+            if isinstance(inner.aggregate, Field):
+                return Subscript(inner.aggregate, inner.index + x.index)
         return x
 
     return map_expr(expr, leaf)
