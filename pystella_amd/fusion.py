@@ -177,17 +177,19 @@ class _StageRedMap:
             return local, rank_shape
         # CPU oracle: reduction of the input state with a lap scratch,
         # then the stage update
-        from itertools import product
-        env2 = dict(env)
-        for lap_name in self.lap_names:
-            f_name = lap_name[len("lap_"):]
-            f = env[f_name]
-            shape = f.shape[:-3] + tuple(rank_shape)
-            lap = torch.zeros(shape, dtype=f.dtype, device=f.device)
-            for s in product(*[range(n) for n in f.shape[:-3]]):
-                self.derivs._apply_lap_cpu(f[s], lap[s])
-            env2[lap_name] = lap
-        local = self.reduction._local_torch(env2, rank_shape)
+        local = None
+        if self.reduction is not None:
+            from itertools import product
+            env2 = dict(env)
+            for lap_name in self.lap_names:
+                f_name = lap_name[len("lap_"):]
+                f = env[f_name]
+                shape = f.shape[:-3] + tuple(rank_shape)
+                lap = torch.zeros(shape, dtype=f.dtype, device=f.device)
+                for s in product(*[range(n) for n in f.shape[:-3]]):
+                    self.derivs._apply_lap_cpu(f[s], lap[s])
+                env2[lap_name] = lap
+            local = self.reduction._local_torch(env2, rank_shape)
         from pystella_amd.backend.torcheval import (
             EvalContext, eval_statements)
         ctx = EvalContext(m.halo_shape, rank_shape)
@@ -355,7 +357,7 @@ class StencilRKStepper:
         key_group = {}         # key field name -> group index
         red_group = 0
         grad_groups = set()    # group indices that use inline grads
-        if reduction is not None and lap_name_list:
+        if lap_name_list:
             by_name = {f.name: f for f in fields}
             # the unknown itself may appear only as a KEY (e.g. the
             # wave equation's h_ij has no bare-field RHS term)
@@ -472,12 +474,14 @@ class StencilRKStepper:
                             out_acc = key
                         rk[out_acc] = key + self_inner._B[stage] * k_new
                         rk_g[gi][out_acc] = rk[out_acc]
-                    if reduction is not None:
+                    if reduction is not None or \
+                            ring_groups is not None:
                         ring = None
                         if ring_groups is not None:
                             ring = [
                                 (rk_g[gi], tmp_g[gi],
                                  red_entries_orig if gi == red_group
+                                 and reduction is not None
                                  else [], fname, nf)
                                 for gi, (fname, nf)
                                 in enumerate(ring_groups)]
@@ -514,7 +518,7 @@ class StencilRKStepper:
         if self._reduction is not None and result is not None:
             local, rank_shape = result
             return self._reduction._combine(local, rank_shape)
-        return result
+        return None
 
     @property
     def tmp_arrays(self):
