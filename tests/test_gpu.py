@@ -767,3 +767,89 @@ def test_classical_rk4_gpu(grid_shape=(16, 16, 16), h=1):
     torch.cuda.synchronize()
     assert (fg - fc).abs().max().item() < 1e-13
     assert (dg - dc).abs().max().item() < 1e-13
+
+
+@requires_gpu
+@pytest.mark.parametrize("h", [1, 3, 4])
+def test_stage_fused_energy_gpu_halo_orders(h, grid_shape=(24, 24, 24)):
+    """Energy-fused ring stage kernel at h=1/3/4 (h=2 covered above)."""
+    from pystella_amd.fusion import StencilRKStepper
+    from pystella_amd.sectors import get_rho_and_p
+    decomp = ps.DomainDecomposition((1, 1, 1), h, rank_shape=grid_shape)
+    dx = (0.3, 0.31, 0.32)
+    dt = 0.01
+    gsize = float(np.prod(grid_shape))
+    sector = ps.ScalarSector(2, potential=lambda f: f[0]**2 / 2)
+    derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=grid_shape)
+    pad = tuple(n + 2 * h for n in grid_shape)
+    torch.manual_seed(41 + h)
+    f0 = torch.rand((2,) + pad, dtype=torch.float64)
+    d0 = torch.rand((2,) + pad, dtype=torch.float64)
+    a = np.ones(1)
+    hub = 0.1 * np.ones(1)
+    cut = (slice(None),) + (slice(h, -h),) * 3
+
+    def run(device):
+        fst = StencilRKStepper(ps.LowStorageRK54, [sector], derivs,
+                               halo_shape=h, rank_shape=grid_shape,
+                               dt=dt, reducers=sector, grid_size=gsize,
+                               callback=get_rho_and_p)
+        arrays = {"f": f0.clone().to(device),
+                  "dfdt": d0.clone().to(device)}
+        arrays["f_next"] = torch.zeros_like(arrays["f"])
+        decomp.share_halos(arrays["f"])
+        es = []
+        for s in range(fst.num_stages):
+            es.append(fst(s, a=a, hubble=hub, **arrays))
+            arrays["f"], arrays["f_next"] = \
+                arrays["f_next"], arrays["f"]
+            decomp.share_halos(arrays["f"])
+        return arrays, es
+
+    ac, ec = run("cpu")
+    ag, eg = run("cuda")
+    torch.cuda.synchronize()
+    for s in range(len(ec)):
+        assert np.allclose(ec[s]["total"], eg[s]["total"],
+                           rtol=1e-12), s
+    assert (ag["f"].cpu()[cut] - ac["f"][cut]).abs().max() < 1e-12
+    assert (ag["dfdt"].cpu()[cut] - ac["dfdt"][cut]).abs().max() < 1e-12
+
+
+@requires_gpu
+def test_multigrid_fas_gpu(n=64, h=1):
+    """Full FAS multigrid solve on GPU (fp32) matches the manufactured
+    solution."""
+    from pystella_amd.field import Field, shift_fields
+    from pystella_amd.multigrid import (
+        FullApproximationScheme, RedBlackIterator, v_cycle)
+    decomp = ps.DomainDecomposition((1, 1, 1), h, rank_shape=(n, n, n))
+    L = 2 * np.pi
+    dx = (L / n,) * 3
+    f = Field("f", offset="h")
+    rho = Field("rho", offset="h")
+    lap = sum(
+        (shift_fields(f, tuple(s * int(m == d) for m in range(3)))
+         - 2 * f
+         + shift_fields(f, tuple(-s * int(m == d) for m in range(3))))
+        for d in range(3) for s in [1]) / var("dx")[0]**2
+    solver = RedBlackIterator(decomp, {f: (lap, rho)}, halo_shape=h,
+                              fixed_parameters=dict(omega=1.0))
+    mg = FullApproximationScheme(solver, halo_shape=h)
+
+    ax = torch.arange(n, dtype=torch.float64) * dx[0]
+    s1 = torch.sin(ax)
+    f_exact = s1[:, None, None] * s1[None, :, None] * s1[None, None, :]
+    pad = (n + 2 * h,) * 3
+    rho_t = torch.zeros(pad, dtype=torch.float64)
+    rho_t[h:-h, h:-h, h:-h] = -3.0 * f_exact
+    rho_t = rho_t.to(torch.float32).cuda()
+    decomp.share_halos(rho_t)
+    ff = torch.zeros(pad, dtype=torch.float32, device="cuda")
+    mg(decomp, dx0=dx, cycle=v_cycle(6, 12, 3), f=ff, rho=rho_t)
+    torch.cuda.synchronize()
+    got = ff[h:-h, h:-h, h:-h].double().cpu()
+    got -= got.mean()
+    want = f_exact - f_exact.mean()
+    rel = (got - want).abs().max().item() / want.abs().max().item()
+    assert rel < 0.05, rel
