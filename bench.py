@@ -203,7 +203,22 @@ def main():
         device_loop = DeviceFriedmannLoop(
             stepper, decomp, expand, grid_size, dt, mpl=mpl)
 
+    prof = None
+    if os.environ.get("PYSTELLA_PROFILE"):
+        from pystella_amd.profiling import Profiler
+        prof = Profiler(enabled=True)
+        step_bytes = 16 * grid_size / max(1, decomp.nranks) * 8 \
+            * stepper.num_stages
+
     def step():
+        nonlocal energy
+        if prof is not None:
+            with prof.region("step", bytes=step_bytes):
+                _step_inner()
+            return
+        _step_inner()
+
+    def _step_inner():
         nonlocal energy
         if device_loop is not None:
             device_loop.step(arrays)
@@ -262,6 +277,10 @@ def main():
 
     ms_per_step = elapsed / p.steps * 1e3
     msites = grid_size * p.steps / elapsed / 1e6
+
+    if prof is not None and decomp.rank == 0:
+        import sys as _sys
+        print(prof.report(), file=_sys.stderr)
 
     if decomp.rank == 0:
         print(json.dumps({
