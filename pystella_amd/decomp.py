@@ -21,12 +21,30 @@ differences, deliberate:
 
 from __future__ import annotations
 
+import logging
 import numbers
+import os
 
 import numpy as np
 import torch
 
 __all__ = ["DomainDecomposition", "init_distributed"]
+
+logger = logging.getLogger(__name__)
+
+# Poor-man's distributed trace to localize hangs, mirroring the
+# reference's debug-level collective tracing with serializing barriers
+# (reference decomp.py:355-363): export PYSTELLA_DEBUG_COLLECTIVES=1
+# (and set logging to DEBUG) to log + barrier around every collective.
+_DEBUG_COLLECTIVES = bool(os.environ.get("PYSTELLA_DEBUG_COLLECTIVES"))
+
+
+def _trace(decomp, what):
+    if _DEBUG_COLLECTIVES:
+        logger.debug("rank %d: entering %s", decomp.rank, what)
+        if decomp.nranks > 1:
+            _dist().barrier()
+        logger.debug("rank %d: %s barrier passed", decomp.rank, what)
 
 
 def _dist():
@@ -204,6 +222,7 @@ class DomainDecomposition:
         ``fx`` (shape ``outer + (nx+2hx, ny+2hy, nz+2hz)``), exchanging
         faces with neighbor ranks along decomposed axes.
         """
+        _trace(self, "share_halos")
         hx, hy, hz = self.halo_shape
         px, py, pz = self.proc_shape
         dim = fx.dim()
@@ -287,6 +306,7 @@ class DomainDecomposition:
 
     def allreduce(self, rank_reduction, op="sum"):
         """All-reduce a scalar, numpy array, or torch tensor."""
+        _trace(self, f"allreduce({op})")
         dist = _dist()
         if self.nranks == 1:
             return rank_reduction
