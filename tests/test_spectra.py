@@ -55,3 +55,42 @@ def test_parseval(grid_shape=(16, 16, 16), L=10.0):
     counts[..., -1] = 1. if grid_shape[2] % 2 == 0 else 2.
     expect = spectra.norm * (counts * np.abs(fk)**2).sum()
     assert abs(total - expect) / expect < 1e-10
+
+
+def _dist_spectra_worker(rank, world_size):
+    """Distributed PowerSpectra over the pencil FFT (gloo, 2 ranks)
+    equals the single-rank result."""
+    grid_shape, L = (16, 16, 16), 10.0
+    decomp = ps.DomainDecomposition((world_size, 1, 1), 0,
+                                    grid_shape=grid_shape)
+    fft = ps.DFT(decomp, grid_shape=grid_shape, dtype=np.float64)
+    dk = tuple(2 * np.pi / L for _ in range(3))
+    spectra = ps.PowerSpectra(decomp, fft, dk, L**3)
+    assert abs(spectra.bin_counts.sum() - np.prod(grid_shape)) < 1e-10
+
+    # deterministic global field, sliced per rank
+    rng = np.random.default_rng(5)
+    full = rng.random(grid_shape)
+    _, start = decomp.get_rank_shape_start(grid_shape)
+    sl = tuple(slice(s, s + n)
+               for s, n in zip(start, decomp.rank_shape))
+    fx = torch.as_tensor(full[sl].copy())
+    spec = spectra(fx, k_power=3)
+
+    # single-rank oracle computed identically on every rank
+    d1 = ps.DomainDecomposition.__new__(ps.DomainDecomposition)
+    d1.proc_shape = (1, 1, 1)
+    d1.rank, d1.nranks = 0, 1
+    d1.rx = d1.ry = d1.rz = 0
+    d1.halo_shape = (0, 0, 0)
+    d1.rank_shape = grid_shape
+    d1.grid_shape = grid_shape
+    fft1 = ps.DFT(d1, grid_shape=grid_shape, dtype=np.float64)
+    spec1 = ps.PowerSpectra(d1, fft1, dk, L**3)(
+        torch.as_tensor(full.copy()), k_power=3)
+    assert np.allclose(spec, spec1, rtol=1e-10), (rank, spec - spec1)
+
+
+def test_distributed_spectra():
+    from tests.conftest import run_distributed
+    run_distributed(_dist_spectra_worker, 2)
