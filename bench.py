@@ -195,8 +195,7 @@ def main():
     expand = ps.Expansion(energy["total"], ps.LowStorageRK54, mpl=mpl)
 
     device_loop = None
-    if fuse_energy and on_gpu and not p.no_device_friedmann \
-            and not p.gws:
+    if fuse_energy and on_gpu and not p.no_device_friedmann:
         # fully device-resident step: stage kernel + partials finish +
         # (RCCL allreduce) + on-device Friedmann ODE — zero host syncs
         from pystella_amd.fusion import DeviceFriedmannLoop

@@ -652,37 +652,49 @@ class DeviceFriedmannLoop:
                 self.stepper._stepper.tmp_arrays = \
                     self.stepper._stepper.get_tmp_arrays_like(**arrays)
             env.update(self.stepper._stepper.tmp_arrays)
-            kern = self._stage_kernel(smap, env)
+            kerns = self._stage_kernels(smap, env)
 
             handles = [self.decomp.share_halos_start(arrays[name])
                        for name in self.stepper.pingpong]
-            interior, slabs = self._regions(kern.rank_shape)
+            interior, slabs = self._regions(kerns[0][0].rank_shape)
             if self._partials is None or \
                     self._boxes != (interior, tuple(slabs)):
                 self._boxes = (interior, tuple(slabs))
-                nblks = [kern.box_nblk(b) for b in (interior, *slabs)]
+                nblks = [kerns[0][0].box_nblk(b)
+                         for b in (interior, *slabs)]
                 self._nblks = nblks
                 self._nblk_tot = sum(nblks)
-                self._partials = torch.empty(
-                    (len(self._red.flat), self._nblk_tot),
-                    dtype=torch.float64, device=f.device)
-            partials = self._partials
+                # one partials buffer per kernel family; only the
+                # reducer family's is finished into sums
+                self._partials = [
+                    torch.empty(
+                        (len(self._red.flat) if has_red else 1,
+                         self._nblk_tot),
+                        dtype=torch.float64, device=f.device)
+                    for _, has_red in kerns]
 
-            kern.launch_box(env, interior, partials, 0, self._nblk_tot)
+            red_partials = None
+            for (kern, has_red), partials in zip(kerns,
+                                                 self._partials):
+                kern.launch_box(env, interior, partials, 0,
+                                self._nblk_tot)
+                if has_red:
+                    red_partials = partials
             for h in handles:
                 h.finish()
             bid0 = self._nblks[0]
             for slab, nb in zip(slabs, self._nblks[1:]):
-                kern.launch_box(env, slab, partials, bid0,
-                                self._nblk_tot)
+                for (kern, _), partials in zip(kerns, self._partials):
+                    kern.launch_box(env, slab, partials, bid0,
+                                    self._nblk_tot)
                 bid0 += nb
 
             if self._fk is None:
                 from pystella_amd.backend.hip import JitFriedmann
                 self._fk = JitFriedmann(
-                    partials.shape[0], self._nblk_tot, self.wt,
+                    red_partials.shape[0], self._nblk_tot, self.wt,
                     self.wp, self.grid_size, mpl=self.mpl)
-            self._fk.finish_sums(partials, self._sums)
+            self._fk.finish_sums(red_partials, self._sums)
             if self.decomp.nranks > 1:
                 dist.all_reduce(self._sums)
             self._fk.step(self._sums, self.state, self._A[s],
@@ -693,29 +705,33 @@ class DeviceFriedmannLoop:
                 env[name] = arrays[name]
                 env[f"{name}_next"] = arrays[f"{name}_next"]
 
-    def _stage_kernel(self, smap, env):
-        if smap.ring is None or len(smap.ring) != 1:
+    def _stage_kernels(self, smap, env):
+        """List of (ring kernel, has_reducers) for this stage, compiled
+        with the device-state scalar map."""
+        if smap.ring is None:
             raise NotImplementedError(
-                "DeviceFriedmannLoop requires a single-family ring "
-                "stepper (use the host fused loop for multi-sector "
-                "runs)")
+                "DeviceFriedmannLoop requires ring-eligible steppers")
         m = smap._map
         rank_shape = m._infer_rank_shape(env)
-        kern = smap._hip_kernel
-        if isinstance(kern, list):
-            kern = None
-        if kern is None or kern.rank_shape != rank_shape or \
-                kern.state_map is None:
+        kerns = smap._hip_kernel
+        ok = (isinstance(kerns, list) and kerns
+              and getattr(kerns[0], "state_map", None) is not None
+              and kerns[0].rank_shape == rank_shape)
+        if not ok:
             from pystella_amd.backend.hip import get_lap_stage_kernel
-            rk_o, tmp_o, red_o, f_name, nf = smap.ring[0]
-            kern = get_lap_stage_kernel(
-                rk_o, tmp_o, red_o, smap._ring_field_args[0], [],
-                m.halo_shape, rank_shape, smap.derivs.dx, nf,
-                f_name=f_name, lap_name=f"lap_{f_name}", name=m.name,
-                state_map={"a": 0, "hubble": 4})
-            smap._hip_kernel = kern
+            kerns = [
+                get_lap_stage_kernel(
+                    rk_o, tmp_o, red_o or [(0.0, "sum")], fargs, [],
+                    m.halo_shape, rank_shape, smap.derivs.dx, nf,
+                    f_name=f_name, lap_name=f"lap_{f_name}",
+                    name=f"{m.name}_{f_name}",
+                    state_map={"a": 0, "hubble": 4})
+                for (rk_o, tmp_o, red_o, f_name, nf), fargs
+                in zip(smap.ring, smap._ring_field_args)]
+            smap._hip_kernel = kerns
             smap._kernel_shape = rank_shape
-        return kern
+        has_red = [bool(r[2]) for r in smap.ring]
+        return list(zip(kerns, has_red))
 
     def read_state(self):
         """Host-side snapshot {a, adot, hubble, energy, pressure} (one
