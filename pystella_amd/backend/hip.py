@@ -830,14 +830,51 @@ class JitLapReduction:
         # Laplacian stencil terms with dx baked in
         inv2 = [1.0 / d / d for d in dx]
         coefs = _LAP_COEFS[h]
+        # periodic=(px,py,pz): wrap that axis's stencil reads in-kernel
+        # (star stencil) so non-decomposed axes need NO halo fill at all
+        px_, py_, pz_ = periodic
+        self.periodic = tuple(periodic)
+        wrap_decls = []
         lap_terms = []
         for s in range(1, h + 1):
             c = coefs[s]
+            if py_:
+                wrap_decls.append(
+                    f"const long ypo{s} = ((j + {s} < NY) ? {s}L "
+                    f": {s}L - NY) * PSZ;")
+                wrap_decls.append(
+                    f"const long ymo{s} = ((j >= {s}) ? -{s}L "
+                    f": NY - {s}L) * PSZ;")
+                yp, ym = f"ypo{s}", f"ymo{s}"
+            else:
+                yp, ym = f"{s}*PSZ", f"-{s}*PSZ"
+            if pz_:
+                wrap_decls.append(
+                    f"const int zpo{s} = (k + {s} < NZ) ? {s} "
+                    f": {s} - NZ;")
+                wrap_decls.append(
+                    f"const int zmo{s} = (k >= {s}) ? -{s} "
+                    f": NZ - {s};")
+                zp, zm = f"zpo{s}", f"zmo{s}"
+            else:
+                zp, zm = f"{s}", f"-{s}"
             lap_terms.append(
                 f"la += {c!r} * ((ring[fld][H+{s}] + ring[fld][H-{s}])"
-                f"*{inv2[0]!r} + (cp[{s}*PSZ] + cp[-{s}*PSZ])*{inv2[1]!r}"
-                f" + (cp[{s}] + cp[-{s}])*{inv2[2]!r});")
+                f"*{inv2[0]!r} + (cp[{yp}] + cp[{ym}])*{inv2[1]!r}"
+                f" + (cp[{zp}] + cp[{zm}])*{inv2[2]!r});")
         lapc0 = coefs[0] * (inv2[0] + inv2[1] + inv2[2])
+        if px_:
+            x_off = ("int xq = i + 2 * H; "
+                     "if (xq >= NX + H) xq -= NX;")
+            ring_load = "ring[fld][2 * H] = fp[(long)xq * sx];"
+            ring_init = ("int xq0 = i0 + p; "
+                         "if (xq0 < H) xq0 += NX; "
+                         "if (xq0 >= NX + H) xq0 -= NX; "
+                         "ring[fld][p] = fp[(long)xq0 * sx];")
+        else:
+            x_off = ""
+            ring_load =                 "ring[fld][2 * H] = fp[(long)(i + 2 * H) * sx];"
+            ring_init = "ring[fld][p] = fp[(long)(i0 + p) * sx];"
 
         # pointer params: stencil field first, then lap (if stored),
         # then the rest
@@ -931,22 +968,25 @@ extern "C" __global__ __launch_bounds__(TBZ * TBY, MINW) void {name}(
     const int i1 = (i0 + XCHUNK < i1b) ? i0 + XCHUNK : i1b;
     if (k < k1b && j < j1b) {{
         const long sx = PSY * PSZ;
+        {wrap_decls}
         double ring[NF][2 * H + 1];
         #pragma unroll
         for (int fld = 0; fld < NF; ++fld) {{
             const double* fp = {fname} + (long)fld * PVOL
                                + (long)(j + H) * PSZ + (k + H);
             #pragma unroll
-            for (int p = 0; p < 2 * H; ++p)
-                ring[fld][p] = fp[(long)(i0 + p) * sx];
+            for (int p = 0; p < 2 * H; ++p) {{
+                {ring_init}
+            }}
         }}
         for (int i = i0; i < i1; ++i) {{
             double lapv[NF];
+            {x_off}
             #pragma unroll
             for (int fld = 0; fld < NF; ++fld) {{
                 const double* fp = {fname} + (long)fld * PVOL
                                    + (long)(j + H) * PSZ + (k + H);
-                ring[fld][2 * H] = fp[(long)(i + 2 * H) * sx];
+                {ring_load}
                 const double* cp = fp + (long)(i + H) * sx;
                 double la = ring[fld][H] * LAPC0;
                 {lap_terms}
@@ -1021,7 +1061,8 @@ class JitLapStage:
     def __init__(self, map_dict, tmp_instructions, entries, field_args,
                  scalar_names, halo, rank_shape, dx, nf, f_name="f",
                  lap_name="lap_f", name="rk_lapstage", tile=None,
-                 nt=True, state_map=None, min_waves=1):
+                 nt=True, state_map=None, min_waves=1,
+                 periodic=(False, False, False)):
         from pystella_amd.derivs import _LAP_COEFS
         from pystella_amd.field import (
             Field, Subscript, iter_exprs, walk_expr)
@@ -1099,14 +1140,51 @@ class JitLapStage:
 
         inv2 = [1.0 / d / d for d in dx]
         coefs = _LAP_COEFS[h]
+        # periodic=(px,py,pz): wrap that axis's stencil reads in-kernel
+        # (star stencil) so non-decomposed axes need NO halo fill at all
+        px_, py_, pz_ = periodic
+        self.periodic = tuple(periodic)
+        wrap_decls = []
         lap_terms = []
         for s in range(1, h + 1):
             c = coefs[s]
+            if py_:
+                wrap_decls.append(
+                    f"const long ypo{s} = ((j + {s} < NY) ? {s}L "
+                    f": {s}L - NY) * PSZ;")
+                wrap_decls.append(
+                    f"const long ymo{s} = ((j >= {s}) ? -{s}L "
+                    f": NY - {s}L) * PSZ;")
+                yp, ym = f"ypo{s}", f"ymo{s}"
+            else:
+                yp, ym = f"{s}*PSZ", f"-{s}*PSZ"
+            if pz_:
+                wrap_decls.append(
+                    f"const int zpo{s} = (k + {s} < NZ) ? {s} "
+                    f": {s} - NZ;")
+                wrap_decls.append(
+                    f"const int zmo{s} = (k >= {s}) ? -{s} "
+                    f": NZ - {s};")
+                zp, zm = f"zpo{s}", f"zmo{s}"
+            else:
+                zp, zm = f"{s}", f"-{s}"
             lap_terms.append(
                 f"la += {c!r} * ((ring[fld][H+{s}] + ring[fld][H-{s}])"
-                f"*{inv2[0]!r} + (cp[{s}*PSZ] + cp[-{s}*PSZ])*{inv2[1]!r}"
-                f" + (cp[{s}] + cp[-{s}])*{inv2[2]!r});")
+                f"*{inv2[0]!r} + (cp[{yp}] + cp[{ym}])*{inv2[1]!r}"
+                f" + (cp[{zp}] + cp[{zm}])*{inv2[2]!r});")
         lapc0 = coefs[0] * (inv2[0] + inv2[1] + inv2[2])
+        if px_:
+            x_off = ("int xq = i + 2 * H; "
+                     "if (xq >= NX + H) xq -= NX;")
+            ring_load = "ring[fld][2 * H] = fp[(long)xq * sx];"
+            ring_init = ("int xq0 = i0 + p; "
+                         "if (xq0 < H) xq0 += NX; "
+                         "if (xq0 >= NX + H) xq0 -= NX; "
+                         "ring[fld][p] = fp[(long)xq0 * sx];")
+        else:
+            x_off = ""
+            ring_load =                 "ring[fld][2 * H] = fp[(long)(i + 2 * H) * sx];"
+            ring_init = "ring[fld][p] = fp[(long)(i0 + p) * sx];"
 
         # pointer params: stencil field first, then every other spatial
         # field referenced by the statements or reducers, then the
@@ -1139,6 +1217,8 @@ class JitLapStage:
             defines=defines, preamble=PREAMBLE, nred=len(entries), nf=nf,
             name=name, params=params, fname=f_name,
             init="\n    ".join(init_lines),
+            wrap_decls="\n        ".join(wrap_decls),
+            x_off=x_off, ring_load=ring_load, ring_init=ring_init,
             lap_terms="\n                ".join(lap_terms),
             body="\n            ".join(lines))
         self.source = src
@@ -1302,11 +1382,12 @@ def get_lap_stage_kernel(map_dict, tmp_instructions, entries, field_args,
                          scalar_names, halo, rank_shape, dx, nf,
                          f_name="f", lap_name="lap_f",
                          name="rk_lapstage", tile=None, nt=True,
-                         state_map=None):
+                         state_map=None, periodic=(False, False, False)):
     return JitLapStage(map_dict, tmp_instructions, entries, field_args,
                        scalar_names, halo, rank_shape, dx, nf,
                        f_name=f_name, lap_name=lap_name, name=name,
-                       tile=tile, nt=nt, state_map=state_map)
+                       tile=tile, nt=nt, state_map=state_map,
+                       periodic=periodic)
 
 
 def get_lap_reduction_kernel(entries, field_args, scalar_names, halo,

@@ -241,7 +241,7 @@ class DomainDecomposition:
                                       self.rz + delta[2])
                 self._exchange_axis(fx, axis, h, (lo_rank, hi_rank))
 
-    def share_halos_start(self, fx):
+    def share_halos_start(self, fx, skip_wrap=False):
         """Overlap-friendly halo exchange for STAR stencils: wraps
         single-rank axes in place immediately (stream-ordered) and posts
         ALL remote-axis face exchanges in one batched non-blocking
@@ -258,9 +258,10 @@ class DomainDecomposition:
         dim = fx.dim()
         ops = []
         fills = []
-        wrap_axes = [ax for ax, (h, p) in enumerate(
-            zip((hx, hy, hz), (px, py, pz))) if h > 0 and p == 1]
-        wrapped_fused = False
+        wrap_axes = [] if skip_wrap else [
+            ax for ax, (h, p) in enumerate(
+                zip((hx, hy, hz), (px, py, pz))) if h > 0 and p == 1]
+        wrapped_fused = skip_wrap
         if wrap_axes and isinstance(fx, torch.Tensor) and fx.is_cuda:
             from pystella_amd.backend.hip import wrap_star
             wrap_star(fx, self.halo_shape, wrap_axes)

@@ -654,7 +654,9 @@ class DeviceFriedmannLoop:
             env.update(self.stepper._stepper.tmp_arrays)
             kerns = self._stage_kernels(smap, env)
 
-            handles = [self.decomp.share_halos_start(arrays[name])
+            skip_wrap = any(kerns[0][0].periodic)
+            handles = [self.decomp.share_halos_start(
+                           arrays[name], skip_wrap=skip_wrap)
                        for name in self.stepper.pingpong]
             interior, slabs = self._regions(kerns[0][0].rank_shape)
             if self._partials is None or \
@@ -705,6 +707,24 @@ class DeviceFriedmannLoop:
                 env[name] = arrays[name]
                 env[f"{name}_next"] = arrays[f"{name}_next"]
 
+    def _periodic_axes(self, smap):
+        """Axes whose stencil reads can wrap in-kernel: non-decomposed
+        axes, and only when no group reads shifted padded fields
+        through the generic codegen (e.g. inlined gradients) — those
+        would still need filled halos."""
+        from pystella_amd.field import collect_fields
+        for rk_o, tmp_o, red_o, _, _ in smap.ring:
+            exprs = (list(tmp_o.values()) + list(rk_o.values())
+                     + [e for e, _ in red_o])
+            for fld in collect_fields(exprs):
+                if fld.is_padded and any(fld.shift):
+                    return (False, False, False)
+        import os
+        if os.environ.get("PYSTELLA_PERIODIC", "1") == "0":
+            return (False, False, False)
+        px, py, pz = self.decomp.proc_shape
+        return (px == 1, py == 1, pz == 1)
+
     def _stage_kernels(self, smap, env):
         """List of (ring kernel, has_reducers) for this stage, compiled
         with the device-state scalar map."""
@@ -719,13 +739,15 @@ class DeviceFriedmannLoop:
               and kerns[0].rank_shape == rank_shape)
         if not ok:
             from pystella_amd.backend.hip import get_lap_stage_kernel
+            periodic = self._periodic_axes(smap)
             kerns = [
                 get_lap_stage_kernel(
                     rk_o, tmp_o, red_o or [(0.0, "sum")], fargs, [],
                     m.halo_shape, rank_shape, smap.derivs.dx, nf,
                     f_name=f_name, lap_name=f"lap_{f_name}",
                     name=f"{m.name}_{f_name}",
-                    state_map={"a": 0, "hubble": 4})
+                    state_map={"a": 0, "hubble": 4},
+                    periodic=periodic)
                 for (rk_o, tmp_o, red_o, f_name, nf), fargs
                 in zip(smap.ring, smap._ring_field_args)]
             smap._hip_kernel = kerns
