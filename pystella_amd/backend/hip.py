@@ -806,7 +806,8 @@ class JitLapReduction:
 
     def __init__(self, entries, field_args, scalar_names, halo, rank_shape,
                  dx, nf, f_name="f", lap_name="lap_f", name="lapred_map",
-                 tile=(64, 4, 64), store_lap=True):
+                 tile=(64, 4, 64), store_lap=True,
+                 periodic=(False, False, False)):
         from pystella_amd.derivs import _LAP_COEFS
         self.rank_shape = tuple(rank_shape)
         self.tile = tile
