@@ -719,8 +719,12 @@ class DeviceFriedmannLoop:
             for fld in collect_fields(exprs):
                 if fld.is_padded and any(fld.shift):
                     return (False, False, False)
+        # measured on MI355X at 512^3: the in-kernel wrap costs ~14
+        # VGPRs (occupancy 5 -> 4 waves/SIMD) which outweighs the saved
+        # wrap launches single-GPU (-2.3%); opt in for launch-bound
+        # small-grid / many-rank regimes with PYSTELLA_PERIODIC=1
         import os
-        if os.environ.get("PYSTELLA_PERIODIC", "1") == "0":
+        if os.environ.get("PYSTELLA_PERIODIC", "0") != "1":
             return (False, False, False)
         px, py, pz = self.decomp.proc_shape
         return (px == 1, py == 1, pz == 1)

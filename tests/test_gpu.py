@@ -375,9 +375,13 @@ def test_scalar_preheating_gws_gpu(tmp_path):
 
 
 @requires_gpu
-def test_device_friedmann_loop_gpu(grid_shape=(32, 32, 32)):
+@pytest.mark.parametrize("periodic", ["0", "1"])
+def test_device_friedmann_loop_gpu(periodic, monkeypatch,
+                                   grid_shape=(32, 32, 32)):
     """Fully device-resident step (stage kernel + on-device Friedmann)
-    matches the host fused loop to fp64 accuracy."""
+    matches the host fused loop to fp64 accuracy — with and without
+    in-kernel periodic stencil reads."""
+    monkeypatch.setenv("PYSTELLA_PERIODIC", periodic)
     from pystella_amd.fusion import (
         DeviceFriedmannLoop, FusedLaplacianReduction, StencilRKStepper)
     from pystella_amd.sectors import get_rho_and_p
