@@ -302,7 +302,7 @@ __global__ __launch_bounds__(BZ * BY) void pd_knl(
 inline int xchunk_size(int nx)
 {
     const char *env = getenv("PYSTELLA_XCHUNK");
-    int c = env ? atoi(env) : 64;
+    int c = env ? atoi(env) : 32;
     if (c <= 0 || c > nx) c = nx;
     return c;
 }
