@@ -731,6 +731,61 @@ def divergence(vec, div, halo=None, dx=None, h=None):
 # HBM traffic per site drops from (f, lap w, f, dfdt, lap r) to
 # (f, dfdt, lap w).
 
+def _lap_stencil_pieces(h, dx, periodic):
+    """Shared codegen for the inline-Laplacian ring kernels: stencil
+    term strings, the center coefficient, and the x-ring load/init
+    forms.  periodic=(px,py,pz) wraps that axis's stencil reads
+    in-kernel (star stencil) so non-decomposed axes need NO halo fill.
+    """
+    from pystella_amd.derivs import _LAP_COEFS
+    inv2 = [1.0 / d / d for d in dx]
+    coefs = _LAP_COEFS[h]
+    px_, py_, pz_ = periodic
+    wrap_decls = []
+    lap_terms = []
+    for s in range(1, h + 1):
+        c = coefs[s]
+        if py_:
+            wrap_decls.append(
+                f"const long ypo{s} = ((j + {s} < NY) ? {s}L "
+                f": {s}L - NY) * PSZ;")
+            wrap_decls.append(
+                f"const long ymo{s} = ((j >= {s}) ? -{s}L "
+                f": NY - {s}L) * PSZ;")
+            yp, ym = f"ypo{s}", f"ymo{s}"
+        else:
+            yp, ym = f"{s}*PSZ", f"-{s}*PSZ"
+        if pz_:
+            wrap_decls.append(
+                f"const int zpo{s} = (k + {s} < NZ) ? {s} "
+                f": {s} - NZ;")
+            wrap_decls.append(
+                f"const int zmo{s} = (k >= {s}) ? -{s} "
+                f": NZ - {s};")
+            zp, zm = f"zpo{s}", f"zmo{s}"
+        else:
+            zp, zm = f"{s}", f"-{s}"
+        lap_terms.append(
+            f"la += {c!r} * ((ring[fld][H+{s}] + ring[fld][H-{s}])"
+            f"*{inv2[0]!r} + (cp[{yp}] + cp[{ym}])*{inv2[1]!r}"
+            f" + (cp[{zp}] + cp[{zm}])*{inv2[2]!r});")
+    lapc0 = coefs[0] * (inv2[0] + inv2[1] + inv2[2])
+    if px_:
+        x_off = ("int xq = i + 2 * H; "
+                 "if (xq >= NX + H) xq -= NX;")
+        ring_load = "ring[fld][2 * H] = fp[(long)xq * sx];"
+        ring_init = ("int xq0 = i0 + p; "
+                     "if (xq0 < H) xq0 += NX; "
+                     "if (xq0 >= NX + H) xq0 -= NX; "
+                     "ring[fld][p] = fp[(long)xq0 * sx];")
+    else:
+        x_off = ""
+        ring_load = "ring[fld][2 * H] = fp[(long)(i + 2 * H) * sx];"
+        ring_init = "ring[fld][p] = fp[(long)(i0 + p) * sx];"
+    return wrap_decls, lap_terms, lapc0, x_off, ring_load, ring_init
+
+
+
 LAPRED_TEMPLATE = """{defines}
 {preamble}
 #define NRED {nred}
@@ -829,53 +884,9 @@ class JitLapReduction:
         combine = "".join(combine_cases) + "0.0"
 
         # Laplacian stencil terms with dx baked in
-        inv2 = [1.0 / d / d for d in dx]
-        coefs = _LAP_COEFS[h]
-        # periodic=(px,py,pz): wrap that axis's stencil reads in-kernel
-        # (star stencil) so non-decomposed axes need NO halo fill at all
-        px_, py_, pz_ = periodic
         self.periodic = tuple(periodic)
-        wrap_decls = []
-        lap_terms = []
-        for s in range(1, h + 1):
-            c = coefs[s]
-            if py_:
-                wrap_decls.append(
-                    f"const long ypo{s} = ((j + {s} < NY) ? {s}L "
-                    f": {s}L - NY) * PSZ;")
-                wrap_decls.append(
-                    f"const long ymo{s} = ((j >= {s}) ? -{s}L "
-                    f": NY - {s}L) * PSZ;")
-                yp, ym = f"ypo{s}", f"ymo{s}"
-            else:
-                yp, ym = f"{s}*PSZ", f"-{s}*PSZ"
-            if pz_:
-                wrap_decls.append(
-                    f"const int zpo{s} = (k + {s} < NZ) ? {s} "
-                    f": {s} - NZ;")
-                wrap_decls.append(
-                    f"const int zmo{s} = (k >= {s}) ? -{s} "
-                    f": NZ - {s};")
-                zp, zm = f"zpo{s}", f"zmo{s}"
-            else:
-                zp, zm = f"{s}", f"-{s}"
-            lap_terms.append(
-                f"la += {c!r} * ((ring[fld][H+{s}] + ring[fld][H-{s}])"
-                f"*{inv2[0]!r} + (cp[{yp}] + cp[{ym}])*{inv2[1]!r}"
-                f" + (cp[{zp}] + cp[{zm}])*{inv2[2]!r});")
-        lapc0 = coefs[0] * (inv2[0] + inv2[1] + inv2[2])
-        if px_:
-            x_off = ("int xq = i + 2 * H; "
-                     "if (xq >= NX + H) xq -= NX;")
-            ring_load = "ring[fld][2 * H] = fp[(long)xq * sx];"
-            ring_init = ("int xq0 = i0 + p; "
-                         "if (xq0 < H) xq0 += NX; "
-                         "if (xq0 >= NX + H) xq0 -= NX; "
-                         "ring[fld][p] = fp[(long)xq0 * sx];")
-        else:
-            x_off = ""
-            ring_load =                 "ring[fld][2 * H] = fp[(long)(i + 2 * H) * sx];"
-            ring_init = "ring[fld][p] = fp[(long)(i0 + p) * sx];"
+        (wrap_decls, lap_terms, lapc0, x_off, ring_load,
+         ring_init) = _lap_stencil_pieces(h, dx, periodic)
 
         # pointer params: stencil field first, then lap (if stored),
         # then the rest
@@ -1139,53 +1150,9 @@ class JitLapStage:
                 lines.append(f"{dst} = {val};")
         combine = "".join(combine_cases) + "0.0"
 
-        inv2 = [1.0 / d / d for d in dx]
-        coefs = _LAP_COEFS[h]
-        # periodic=(px,py,pz): wrap that axis's stencil reads in-kernel
-        # (star stencil) so non-decomposed axes need NO halo fill at all
-        px_, py_, pz_ = periodic
         self.periodic = tuple(periodic)
-        wrap_decls = []
-        lap_terms = []
-        for s in range(1, h + 1):
-            c = coefs[s]
-            if py_:
-                wrap_decls.append(
-                    f"const long ypo{s} = ((j + {s} < NY) ? {s}L "
-                    f": {s}L - NY) * PSZ;")
-                wrap_decls.append(
-                    f"const long ymo{s} = ((j >= {s}) ? -{s}L "
-                    f": NY - {s}L) * PSZ;")
-                yp, ym = f"ypo{s}", f"ymo{s}"
-            else:
-                yp, ym = f"{s}*PSZ", f"-{s}*PSZ"
-            if pz_:
-                wrap_decls.append(
-                    f"const int zpo{s} = (k + {s} < NZ) ? {s} "
-                    f": {s} - NZ;")
-                wrap_decls.append(
-                    f"const int zmo{s} = (k >= {s}) ? -{s} "
-                    f": NZ - {s};")
-                zp, zm = f"zpo{s}", f"zmo{s}"
-            else:
-                zp, zm = f"{s}", f"-{s}"
-            lap_terms.append(
-                f"la += {c!r} * ((ring[fld][H+{s}] + ring[fld][H-{s}])"
-                f"*{inv2[0]!r} + (cp[{yp}] + cp[{ym}])*{inv2[1]!r}"
-                f" + (cp[{zp}] + cp[{zm}])*{inv2[2]!r});")
-        lapc0 = coefs[0] * (inv2[0] + inv2[1] + inv2[2])
-        if px_:
-            x_off = ("int xq = i + 2 * H; "
-                     "if (xq >= NX + H) xq -= NX;")
-            ring_load = "ring[fld][2 * H] = fp[(long)xq * sx];"
-            ring_init = ("int xq0 = i0 + p; "
-                         "if (xq0 < H) xq0 += NX; "
-                         "if (xq0 >= NX + H) xq0 -= NX; "
-                         "ring[fld][p] = fp[(long)xq0 * sx];")
-        else:
-            x_off = ""
-            ring_load =                 "ring[fld][2 * H] = fp[(long)(i + 2 * H) * sx];"
-            ring_init = "ring[fld][p] = fp[(long)(i0 + p) * sx];"
+        (wrap_decls, lap_terms, lapc0, x_off, ring_load,
+         ring_init) = _lap_stencil_pieces(h, dx, periodic)
 
         # pointer params: stencil field first, then every other spatial
         # field referenced by the statements or reducers, then the
