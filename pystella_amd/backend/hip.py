@@ -1081,6 +1081,9 @@ class JitLapStage:
         if tile is None:
             tile = self.pick_tile(rank_shape)
         self.rank_shape = tuple(rank_shape)
+        # dynamic-LDS ballast at launch: caps waves/CU without kernel
+        # changes (occupancy throttling for cache-thrash regimes)
+        self.shmem = 0
         self.tile = tile
         self.entries = entries
         h = max(halo) if isinstance(halo, (tuple, list)) else halo
@@ -1224,7 +1227,7 @@ class JitLapStage:
         grid, ints = self._box_geometry(box)
         doubles = [_resolve_scalar(env, k) for k in self.scalar_keys]
         ext().jit_launch(self.key, grid[0], grid[1], grid[2],
-                         self.block, 1, 1, 0, _stream(),
+                         self.block, 1, 1, self.shmem, _stream(),
                          ptrs + [partials.data_ptr()],
                          ints + [nblk_tot, bid0], doubles)
 
