@@ -457,10 +457,19 @@ class StencilRKStepper:
                         tmp_g[gi][rhs_name] = rhs_dict_orig[key]
                         # keep the updated k in a register: one load and
                         # one store of the k array per site, and stores
-                        # are never read back (safe for nontemporal)
+                        # are never read back (safe for nontemporal).
+                        # For large families (GW hij), stage 0's elided
+                        # A*k term produces a measurably SLOWER kernel
+                        # (18 vs 13 ms at 512^3) — keep the term with a
+                        # runtime-zero coefficient so the compiled form
+                        # (incl. the latency-hiding k preloads) matches
+                        # the other stages.
+                        coefA = self_inner._A[stage]
+                        if (stage == 0 and ring_groups is not None
+                                and ring_groups[gi][1] >= 4):
+                            coefA = var("rk_a0")
                         k_new = var(f"knew_{i}")
-                        tmp[k_new] = (self_inner._A[stage] * k_acc
-                                      + dtv * rhs_name)
+                        tmp[k_new] = coefA * k_acc + dtv * rhs_name
                         tmp_g[gi][k_new] = tmp[k_new]
                         rk[k_acc] = k_new
                         rk_g[gi][k_acc] = k_new
@@ -485,6 +494,8 @@ class StencilRKStepper:
                                  else [], fname, nf)
                                 for gi, (fname, nf)
                                 in enumerate(ring_groups)]
+                        fp = dict(fixed_parameters or {})
+                        fp.setdefault("rk_a0", 0.0)
                         steps.append(_StageRedMap(
                             rk, tmp_instructions=tmp,
                             red_entries=red_entries,
@@ -493,7 +504,7 @@ class StencilRKStepper:
                             halo_shape=self_inner.halo_shape,
                             rank_shape=self_inner.rank_shape,
                             name=f"rk_stage_red{stage}",
-                            fixed_parameters=fixed_parameters, **kw))
+                            fixed_parameters=fp, **kw))
                     else:
                         steps.append(ElementWiseMap(
                             rk, tmp_instructions=tmp,
@@ -644,6 +655,7 @@ class DeviceFriedmannLoop:
         env = dict(arrays)
         env["state"] = self.state
         env["dt"] = self.dt
+        env["rk_a0"] = 0.0
         if extra_scalars:
             env.update(extra_scalars)
         for s in range(self.num_stages):
