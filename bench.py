@@ -40,6 +40,10 @@ def main():
     parser.add_argument("--grid", type=int, default=512)
     parser.add_argument("--halo", type=int, default=2)
     parser.add_argument("--device", default=None)
+    parser.add_argument("--stepper", default="LowStorageRK54",
+                        help="low-storage stepper class name (any "
+                             "2N-storage tableau: LowStorageRK54/144/"
+                             "134/124/3Williamson/...)")
     parser.add_argument("--no-fuse", action="store_true",
                         help="reference-structure loop (separate lap "
                              "array + unfused stage kernels)")
@@ -97,6 +101,7 @@ def main():
         phi, chi = f[0], f[1]
         return (mphi**2 / 2 * phi**2 + gsq / 2 * phi**2 * chi**2) / mphi**2
 
+    Stepper = getattr(ps, p.stepper)
     sector = ps.ScalarSector(nscalars, potential=potential)
     sectors = [sector]
     fuse_energy_pre = not (p.no_fuse or p.no_fuse_energy)
@@ -115,14 +120,14 @@ def main():
         FusedLaplacianReduction, StencilRKStepper)
     fuse_energy = not (p.no_fuse or p.no_fuse_energy)
     if p.no_fuse:
-        stepper = ps.LowStorageRK54(sectors, halo_shape=h,
+        stepper = Stepper(sectors, halo_shape=h,
                                     rank_shape=rank_shape, dt=dt)
     elif fuse_energy:
         # fully fused MI355X structure: one kernel per RK stage that
         # evaluates the Laplacian inline (ping-pong f), updates the
         # unknowns AND reduces the input-state energy — no separate
         # energy pass in the hot loop at all
-        stepper = StencilRKStepper(ps.LowStorageRK54, sectors, derivs,
+        stepper = StencilRKStepper(Stepper, sectors, derivs,
                                    halo_shape=h, rank_shape=rank_shape,
                                    dt=dt, reducers=sector,
                                    grid_size=grid_size,
@@ -131,7 +136,7 @@ def main():
     else:
         # stage kernels evaluate the Laplacian inline (ping-pong f);
         # the energy reduction is a separate fused lap+reduce kernel
-        stepper = StencilRKStepper(ps.LowStorageRK54, sectors, derivs,
+        stepper = StencilRKStepper(Stepper, sectors, derivs,
                                    halo_shape=h, rank_shape=rank_shape,
                                    dt=dt)
     reduce_energy = FusedLaplacianReduction(
@@ -192,7 +197,7 @@ def main():
         return reduce_energy(a=np.array(a), **kw)
 
     energy = compute_energy(1.)
-    expand = ps.Expansion(energy["total"], ps.LowStorageRK54, mpl=mpl)
+    expand = ps.Expansion(energy["total"], Stepper, mpl=mpl)
 
     device_loop = None
     if fuse_energy and on_gpu and not p.no_device_friedmann:
@@ -301,7 +306,8 @@ def main():
                 "grid_shape": list(grid_shape),
                 "halo": h,
                 "nscalars": nscalars,
-                "stepper": "LowStorageRK54 (4th order, 5 stages)",
+                "stepper": f"{p.stepper} "
+                           f"({stepper.num_stages} stages)",
                 "global_batch": None,
                 "seq_len": None,
                 "parallelism": f"decomp3d{list(proc_shape)}",
