@@ -719,7 +719,7 @@ class DeviceFriedmannLoop:
                 env[name] = arrays[name]
                 env[f"{name}_next"] = arrays[f"{name}_next"]
 
-    def _periodic_axes(self, smap):
+    def _periodic_axes(self, smap, rank_shape):
         """Axes whose stencil reads can wrap in-kernel: non-decomposed
         axes, and only when no group reads shifted padded fields
         through the generic codegen (e.g. inlined gradients) — those
@@ -731,12 +731,16 @@ class DeviceFriedmannLoop:
             for fld in collect_fields(exprs):
                 if fld.is_padded and any(fld.shift):
                     return (False, False, False)
-        # measured on MI355X at 512^3: the in-kernel wrap costs ~14
-        # VGPRs (occupancy 5 -> 4 waves/SIMD) which outweighs the saved
-        # wrap launches single-GPU (-2.3%); opt in for launch-bound
-        # small-grid / many-rank regimes with PYSTELLA_PERIODIC=1
+        # measured on MI355X: at 512^3 the in-kernel wrap costs ~14
+        # VGPRs (occupancy 5 -> 4 waves/SIMD) and loses 2.3%, but at
+        # 128^3 (launch-bound, the strong-scaling regime) it WINS 12%
+        # (0.437 -> 0.391 ms/step).  Default: on for small per-rank
+        # grids; PYSTELLA_PERIODIC=1/0 forces either way.
         import os
-        if os.environ.get("PYSTELLA_PERIODIC", "0") != "1":
+        force = os.environ.get("PYSTELLA_PERIODIC")
+        if force == "0":
+            return (False, False, False)
+        if force != "1" and int(np.prod(rank_shape)) > 4_000_000:
             return (False, False, False)
         px, py, pz = self.decomp.proc_shape
         return (px == 1, py == 1, pz == 1)
@@ -755,7 +759,7 @@ class DeviceFriedmannLoop:
               and kerns[0].rank_shape == rank_shape)
         if not ok:
             from pystella_amd.backend.hip import get_lap_stage_kernel
-            periodic = self._periodic_axes(smap)
+            periodic = self._periodic_axes(smap, rank_shape)
             kerns = [
                 get_lap_stage_kernel(
                     rk_o, tmp_o, red_o or [(0.0, "sum")], fargs, [],
