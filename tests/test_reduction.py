@@ -331,3 +331,25 @@ def test_device_loop_region_partition():
         for (i0, i1, j0, j1, k0, k1) in [interior] + slabs:
             count[i0:i1, j0:j1, k0:k1] += 1
         assert (count == 1).all(), (proc_shape, count.min(), count.max())
+
+
+def test_field_statistics_min_max(grid_shape=(12, 12, 12)):
+    """FieldStatistics with max_min=True (reference reduction.py:284-302)."""
+    h = 1
+    decomp = ps.DomainDecomposition((1, 1, 1), h, rank_shape=grid_shape)
+    stats = ps.FieldStatistics(decomp, h, rank_shape=grid_shape,
+                               max_min=True)
+    torch.manual_seed(77)
+    pad = tuple(n + 2 * h for n in grid_shape)
+    f = torch.rand((2,) + pad, dtype=torch.float64) - 0.3
+    out = stats(f)
+    cut = (slice(None),) + (slice(h, -h),) * 3
+    fi = f[cut]
+    for i in range(2):
+        assert np.isclose(out["mean"][i], fi[i].mean().item())
+        assert np.isclose(out["variance"][i], fi[i].var(
+            unbiased=False).item(), rtol=1e-10)
+        assert np.isclose(out["max"][i], fi[i].max().item())
+        assert np.isclose(out["min"][i], fi[i].min().item())
+        assert np.isclose(out["abs_max"][i], fi[i].abs().max().item())
+        assert np.isclose(out["abs_min"][i], fi[i].abs().min().item())
