@@ -37,7 +37,11 @@ __global__ __launch_bounds__(256) void tt_project_mfma(
     int ny, int nz, long vol)
 {
     const int lane = threadIdx.x & 63;
-    const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    // 2-D grid: HSA grid_size per dimension is uint32 WORK-ITEMS, so a
+    // 1-D launch overflows for k-volumes above ~2.7e8 sites (>= 1024^3
+    // r2c); blockIdx.y carries the high part.
+    const long block = (long)blockIdx.y * gridDim.x + blockIdx.x;
+    const long wave = (block * blockDim.x + threadIdx.x) >> 6;
     const int blk = (lane & 15) >> 2;
     long site = wave * 4 + blk;
     const bool active = site < vol;
@@ -117,7 +121,10 @@ extern "C" void launch_tt_project_mfma(
     const long waves = (vol + 3) / 4;
     const long threads = waves * 64;
     const int block = 256;
-    const long grid = (threads + block - 1) / block;
-    hipLaunchKernelGGL(tt_project_mfma, dim3((uint32_t)grid), dim3(block),
-                       0, stream, hij, out, kx, ky, kz, ny, nz, vol);
+    const long nblocks = (threads + block - 1) / block;
+    const long gx = nblocks < 32768 ? nblocks : 32768;
+    const long gy = (nblocks + gx - 1) / gx;
+    hipLaunchKernelGGL(tt_project_mfma, dim3((uint32_t)gx, (uint32_t)gy),
+                       dim3(block), 0, stream, hij, out, kx, ky, kz, ny,
+                       nz, vol);
 }
