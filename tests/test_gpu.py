@@ -488,7 +488,8 @@ def test_relax_gpu_matches_cpu(tdtype, n=32, h=1):
 
 
 @requires_gpu
-def test_tt_projection_mfma_vs_oracle(grid_shape=(16, 16, 16)):
+@pytest.mark.parametrize("grid_shape", [(16, 16, 16), (64, 48, 32)])
+def test_tt_projection_mfma_vs_oracle(grid_shape):
     """Transverse-traceless projection on the f64 matrix cores
     (csrc/tt_mfma.hip) vs the torch oracle: transversality, traceless-
     ness and exact agreement."""
