@@ -194,3 +194,15 @@ def test_overlap_halo_exchange_x():
 
 def test_overlap_halo_exchange_z():
     run_distributed(_overlap_halo_worker, 2, args=((1, 1, 2), (8, 8, 8), 2))
+
+
+def test_halo_exchange_2d_pencil():
+    """4 ranks, (2,2,1) pencil — the reference CI's mpirun -np 4
+    --proc_shape 2,2,1 configuration (ci.yml:97-99)."""
+    run_distributed(_halo_worker, 4, args=((2, 2, 1), (8, 8, 8), 2))
+
+
+def test_halo_exchange_3d():
+    """8 ranks, full 3-D (2,2,2) decomposition (beyond the reference,
+    which caps at 2-D)."""
+    run_distributed(_halo_worker, 8, args=((2, 2, 2), (8, 8, 8), 1))
