@@ -151,3 +151,32 @@ def test_bench_distributed_gws_cpu(tmp_path):
     line = [ln for ln in logs.splitlines() if ln.startswith("{")][-1]
     d = json.loads(line)
     assert d["config"]["model"] == "scalar_preheating+gw"
+
+
+def test_bench_8rank_cpu(tmp_path):
+    """8-rank (2,2,2) bench on gloo — the exact topology of the
+    driver's N=8 strong-scaling run."""
+    import glob
+    import json
+    import subprocess
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [sys.executable, "-m", "torch.distributed.run",
+           "--standalone", "--local-addr", "127.0.0.1",
+           "--nnodes=1", "--nproc-per-node", "8",
+           "--redirects", "3", "--log-dir", str(tmp_path / "trlogs"),
+           os.path.join(repo, "bench.py"),
+           "--gpus", "8", "--steps", "1", "--warmup", "0",
+           "--grid", "16", "--device", "cpu"]
+    env = {k: v for k, v in os.environ.items()
+           if k not in ("RANK", "LOCAL_RANK", "WORLD_SIZE",
+                        "MASTER_ADDR", "MASTER_PORT", "LOCAL_WORLD_SIZE",
+                        "GROUP_RANK", "TORCHELASTIC_RUN_ID")}
+    out = subprocess.run(cmd, capture_output=True, text=True,
+                         timeout=600, cwd=repo, env=env)
+    logs = "\n".join(open(f).read() for f in glob.glob(
+        str(tmp_path / "trlogs" / "**" / "*.log"), recursive=True))
+    assert out.returncode == 0, (out.stderr[-1500:], logs[-1500:])
+    line = [ln for ln in logs.splitlines() if ln.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["config"]["parallelism"] == "decomp3d[2, 2, 2]"
+    assert d["value"] > 0
