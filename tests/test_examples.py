@@ -161,7 +161,7 @@ def test_bench_8rank_cpu(tmp_path):
     import subprocess
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     cmd = [sys.executable, "-m", "torch.distributed.run",
-           "--standalone", "--local-addr", "127.0.0.1",
+           "--master-addr", "127.0.0.1", "--master-port", "29833",
            "--nnodes=1", "--nproc-per-node", "8",
            "--redirects", "3", "--log-dir", str(tmp_path / "trlogs"),
            os.path.join(repo, "bench.py"),
