@@ -70,6 +70,12 @@ def main():
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     if world_size > 1:
         ps.init_distributed()
+    if p.gpus > 1 and world_size != p.gpus:
+        raise SystemExit(
+            f"--gpus {p.gpus} requires a torchrun launch with "
+            f"WORLD_SIZE={p.gpus} (got WORLD_SIZE={world_size}); a "
+            f"single-process run would do 1-GPU work but report "
+            f"n_gpus={p.gpus}")
     n_gpus = max(p.gpus, world_size)
 
     if p.device is not None:

@@ -57,9 +57,13 @@ class Codegen:
             return self._scalar_index[key]
         c_name = name if not idx else name + "_" + "_".join(map(str, idx))
         c_name = "s_" + c_name
-        self._scalar_index[key] = c_name
+        # cache the real()-wrapped form so repeated references evaluate
+        # in the kernel's real type consistently (scalars are passed as
+        # double args; in float kernels an unwrapped later use would
+        # silently promote the expression to double)
+        self._scalar_index[key] = f"real({c_name})"
         self.scalars.append((c_name, key))
-        return f"real({c_name})"
+        return self._scalar_index[key]
 
     # ------------------------------------------------------------------
     def field_access(self, f: Field, outer_idx):

@@ -204,6 +204,7 @@ class JitStageReduction:
                  tile=(64, 4, 64)):
         self.rank_shape = tuple(rank_shape)
         self.tile = tile
+        self.dtype = torch.float64      # fp64-only kernel (double* params)
         self.entries = entries
         self.field_args = [fa for fa in field_args if fa.spatial]
         nred = len(entries)
@@ -251,8 +252,15 @@ class JitStageReduction:
     def __call__(self, env):
         dev = None
         ptrs = []
+        want = getattr(self, "dtype", None)
         for fa in self.field_args:
             t = _check_tensor(fa.name, env[fa.name])
+            if want is None:
+                want = t.dtype
+            elif t.dtype != want:
+                raise TypeError(
+                    f"reduction argument '{fa.name}' has dtype "
+                    f"{t.dtype}, kernel expects {want}")
             dev = t.device
             ptrs.append(t.data_ptr())
         nred = len(self.entries)
@@ -485,8 +493,15 @@ class JitReduction:
     def __call__(self, env):
         dev = None
         ptrs = []
+        want = getattr(self, "dtype", None)
         for fa in self.field_args:
             t = _check_tensor(fa.name, env[fa.name])
+            if want is None:
+                want = t.dtype
+            elif t.dtype != want:
+                raise TypeError(
+                    f"reduction argument '{fa.name}' has dtype "
+                    f"{t.dtype}, kernel expects {want}")
             dev = t.device
             ptrs.append(t.data_ptr())
         nred = len(self.entries)

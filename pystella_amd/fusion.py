@@ -589,6 +589,13 @@ class DeviceFriedmannLoop:
             self.wt[r] = float(np.asarray(out["total"]).reshape(-1)[0])
             self.wp[r] = float(
                 np.asarray(out["pressure"]).reshape(-1)[0])
+            # JitFriedmann divides EVERY entry by grid_size (avg
+            # semantics); 'sum' reducers must not be averaged, so bake
+            # the compensating factor into the weights (matches the
+            # host path, Reduction._combine, which only divides 'avg')
+            if red.flat[r][3] == "sum":
+                self.wt[r] *= self.grid_size
+                self.wp[r] *= self.grid_size
 
         # device state [a, adot, k_a, k_adot, hubble, energy, pressure]
         self.state = None

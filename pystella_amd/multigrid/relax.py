@@ -188,7 +188,14 @@ class RedBlackIterator(RelaxationBase):
     def __call__(self, decomp, queue=None, iterations=100, **kwargs):
         kwargs.pop("solve_constraint", None)
         if "rb_off" not in kwargs:
-            kwargs["rb_off"] = 0.0
+            # checkerboard color is of the GLOBAL site: offset the
+            # rank-local (i+j+k) parity by the parity of this rank's
+            # global start so colors agree across rank seams
+            rb_off = 0.0
+            if decomp is not None and decomp.grid_shape is not None:
+                _, start = decomp.get_rank_shape_start(decomp.grid_shape)
+                rb_off = float(sum(start) % 2)
+            kwargs["rb_off"] = rb_off
         for _ in range(iterations):
             for stepper in self.color_steppers:
                 stepper(**{k: v for k, v in kwargs.items()

@@ -160,10 +160,17 @@ class FullApproximationScheme:
             if i not in self.dx:
                 self.dx[i] = np.array(self.dx[i - 1] * 2)
             if i not in self.decomp:
-                ng_2 = tuple(n // 2 for n in self.decomp[i - 1].rank_shape)
+                # MG levels require per-rank shapes that halve evenly, so
+                # every rank holds the same ng_2; construct the level
+                # decomp from the (even-division) global shape so rank
+                # global-start offsets are known (RBGS checkerboard
+                # parity across seams needs them)
+                prev = self.decomp[i - 1]
+                ng_2 = tuple(n // 2 for n in prev.rank_shape)
                 self.decomp[i] = DomainDecomposition(
-                    self.decomp[i - 1].proc_shape, self.halo_shape,
-                    rank_shape=ng_2)
+                    prev.proc_shape, self.halo_shape,
+                    grid_shape=tuple(n * p for n, p in
+                                     zip(ng_2, prev.proc_shape)))
             if i not in self.unknowns:
                 self.unknowns[i] = self.coarse_level_like(
                     self.unknowns[i - 1])
