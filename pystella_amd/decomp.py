@@ -88,9 +88,10 @@ def get_size_start(N, size, rank):
 class _HaloHandle:
     """Pending overlapped halo exchange (see share_halos_start)."""
 
-    def __init__(self, works, fills):
+    def __init__(self, works, fills, release=None):
         self.works = works
         self.fills = fills
+        self._release = release
         self._done = False
 
     def finish(self):
@@ -100,6 +101,8 @@ class _HaloHandle:
             w.wait()
         for dst, src in self.fills:
             dst.copy_(src)
+        if self._release is not None:
+            self._release()
         self._done = True
 
 
@@ -143,6 +146,29 @@ class DomainDecomposition:
             rank_shape, _ = self.get_rank_shape_start(grid_shape)
         self.rank_shape = tuple(rank_shape) if rank_shape is not None else None
         self.grid_shape = tuple(grid_shape) if grid_shape is not None else None
+
+        # persistent communication buffer pool, keyed by
+        # (shape, dtype, device): avoids a contiguous()+empty_like
+        # allocator round trip per face per field per RK stage
+        # (the reference caches its comm buffers too, decomp.py:339-349)
+        self._buf_pool = {}
+
+    def _acquire_bufs(self, shape, dtype, device, n=4):
+        """Check out ``n`` persistent contiguous buffers of the given
+        face shape (returned to the pool via :meth:`_release_bufs`)."""
+        key = (tuple(shape), dtype, str(device))
+        pool = self._buf_pool.setdefault(key, [])
+        out = []
+        for _ in range(n):
+            if pool:
+                out.append(pool.pop())
+            else:
+                out.append(torch.empty(shape, dtype=dtype, device=device))
+        return out
+
+    def _release_bufs(self, shape, dtype, device, bufs):
+        key = (tuple(shape), dtype, str(device))
+        self._buf_pool.setdefault(key, []).extend(bufs)
 
     # -- topology -----------------------------------------------------------
 
@@ -194,10 +220,11 @@ class DomainDecomposition:
             sl[axis] = slice(lo, lo + extent)
             return fx[tuple(sl)]
 
-        send_lo = face(h, h).contiguous()          # my low interior face
-        send_hi = face(n, h).contiguous()          # my high interior face
-        recv_lo = torch.empty_like(send_lo)        # fills halo [0:h]
-        recv_hi = torch.empty_like(send_hi)        # fills halo [n+h:n+2h]
+        fshape = tuple(face(h, h).shape)
+        send_lo, send_hi, recv_lo, recv_hi = self._acquire_bufs(
+            fshape, fx.dtype, fx.device)
+        send_lo.copy_(face(h, h))                  # my low interior face
+        send_hi.copy_(face(n, h))                  # my high interior face
         lo_rank, hi_rank = neighbors
 
         # Pairing convention (also correct when lo_rank == hi_rank, e.g.
@@ -216,6 +243,8 @@ class DomainDecomposition:
 
         face(0, h).copy_(recv_lo)
         face(n + h, h).copy_(recv_hi)
+        self._release_bufs(fshape, fx.dtype, fx.device,
+                           (send_lo, send_hi, recv_lo, recv_hi))
 
     def share_halos(self, fx):
         """Impose periodic boundary conditions on the halo padding of
@@ -258,6 +287,7 @@ class DomainDecomposition:
         dim = fx.dim()
         ops = []
         fills = []
+        checked_out = []
         wrap_axes = [] if skip_wrap else [
             ax for ax, (h, p) in enumerate(
                 zip((hx, hy, hz), (px, py, pz))) if h > 0 and p == 1]
@@ -281,10 +311,13 @@ class DomainDecomposition:
                 sl[axis] = slice(lo, lo + extent)
                 return fx[tuple(sl)]
 
-            send_lo = face(h, h).contiguous()
-            send_hi = face(n, h).contiguous()
-            recv_lo = torch.empty_like(send_lo)
-            recv_hi = torch.empty_like(send_hi)
+            fshape = tuple(face(h, h).shape)
+            send_lo, send_hi, recv_lo, recv_hi = self._acquire_bufs(
+                fshape, fx.dtype, fx.device)
+            send_lo.copy_(face(h, h))
+            send_hi.copy_(face(n, h))
+            checked_out.append((fshape, fx.dtype, fx.device,
+                                (send_lo, send_hi, recv_lo, recv_hi)))
             delta = [0, 0, 0]
             delta[ax_rel] = 1
             lo_rank = self.rankID(self.rx - delta[0], self.ry - delta[1],
@@ -301,12 +334,29 @@ class DomainDecomposition:
             fills.append((face(n + h, h), recv_hi))
 
         works = dist.batch_isend_irecv(ops) if ops else []
-        return _HaloHandle(works, fills)
+
+        def release():
+            for shape, dtype, device, bufs in checked_out:
+                self._release_bufs(shape, dtype, device, bufs)
+
+        return _HaloHandle(works, fills, release=release)
 
     # -- collectives --------------------------------------------------------
 
+    def _comm_backend_is_nccl(self):
+        dist = _dist()
+        try:
+            return "nccl" in str(dist.get_backend())
+        except (RuntimeError, ValueError):
+            return False
+
     def allreduce(self, rank_reduction, op="sum"):
-        """All-reduce a scalar, numpy array, or torch tensor."""
+        """All-reduce a scalar, numpy array, or torch tensor.
+
+        RCCL (the ``"nccl"`` backend) only reduces device tensors, so
+        host-side scalars/arrays are staged through the GPU when that
+        is the communicator backend.
+        """
         _trace(self, f"allreduce({op})")
         dist = _dist()
         if self.nranks == 1:
@@ -316,18 +366,21 @@ class DomainDecomposition:
                   "min": dist.ReduceOp.MIN, "prod": dist.ReduceOp.PRODUCT,
                   }[op]
 
-        if isinstance(rank_reduction, torch.Tensor):
-            t = rank_reduction.clone()
+        def _reduce(t):
+            if not t.is_cuda and self._comm_backend_is_nccl():
+                d = t.cuda()
+                dist.all_reduce(d, op=red_op)
+                return d.cpu()
             dist.all_reduce(t, op=red_op)
             return t
+
+        if isinstance(rank_reduction, torch.Tensor):
+            return _reduce(rank_reduction.clone())
         if isinstance(rank_reduction, np.ndarray):
-            t = torch.from_numpy(rank_reduction.copy())
-            dist.all_reduce(t, op=red_op)
-            return t.numpy()
+            return _reduce(torch.from_numpy(rank_reduction.copy())).numpy()
         if isinstance(rank_reduction, numbers.Number):
             t = torch.tensor([rank_reduction], dtype=torch.float64)
-            dist.all_reduce(t, op=red_op)
-            return t.item()
+            return _reduce(t).item()
         raise TypeError(f"cannot allreduce {type(rank_reduction)}")
 
     def bcast(self, x, root=0):

@@ -16,6 +16,8 @@ composes :class:`~pystella_amd.FiniteDifferencer` +
 
 from __future__ import annotations
 
+import os
+
 import numpy as np
 import torch
 
@@ -674,9 +676,17 @@ class DeviceFriedmannLoop:
             kerns = self._stage_kernels(smap, env)
 
             skip_wrap = any(kerns[0][0].periodic)
-            handles = [self.decomp.share_halos_start(
-                           arrays[name], skip_wrap=skip_wrap)
-                       for name in self.stepper.pingpong]
+            # PYSTELLA_NO_OVERLAP=1: safety valve for real-xGMI bring-up
+            # — sequential per-axis share_halos (no concurrent batched
+            # group, corners propagated) instead of the overlapped path
+            if os.environ.get("PYSTELLA_NO_OVERLAP") == "1":
+                for name in self.stepper.pingpong:
+                    self.decomp.share_halos(arrays[name])
+                handles = []
+            else:
+                handles = [self.decomp.share_halos_start(
+                               arrays[name], skip_wrap=skip_wrap)
+                           for name in self.stepper.pingpong]
             interior, slabs = self._regions(kerns[0][0].rank_shape)
             if self._partials is None or \
                     self._boxes != (interior, tuple(slabs)):
