@@ -176,10 +176,26 @@ class TorchDFT(BaseDFT):
 
 
 class PencilDFT(BaseDFT):
-    """Distributed pencil FFT over a (px, py, 1) processor grid.
+    """Distributed pencil FFT over any (px, py, pz) processor grid.
 
     k-space layout: kx full on every rank; ky split over px (this
-    rank's rx); kz split over py (this rank's ry).
+    rank's rx); kz split over the py·pz "row" dimension (position
+    q = ry·pz + rz).  For pz == 1 this is the classic 2-D pencil
+    (ky over px, kz over py); for pz > 1 an extra z-join transpose
+    within each (rx, ry) sub-group precedes the z-FFT, so the full 3-D
+    decomposition — including the driver's (2, 2, 2) N=8 topology —
+    supports every FFT observable.  (The reference's ``pDFT`` covers
+    only its 2-D decomposition space, dft.py:391-404, and runs on the
+    host through mpi4py-fft; this is an in-HBM RCCL design.)
+
+    Transpose chain (forward):
+
+        [pz > 1] z-group all-to-all: split y by pz, join z
+        rfft(z)
+        [py·pz > 1] row all-to-all: split kz by py·pz, join y
+        fft(y)
+        [px > 1] col all-to-all: split y by px, join x
+        fft(x)
     """
 
     def __init__(self, decomp, grid_shape, dtype=np.float64, device="cpu"):
@@ -190,10 +206,9 @@ class PencilDFT(BaseDFT):
         self.is_real = self.dtype.kind == "f"
         self.device = torch.device(device)
         px, py, pz = decomp.proc_shape
-        if pz != 1:
-            raise NotImplementedError(
-                "pencil FFT requires proc_shape[2] == 1")
-        self.px, self.py = px, py
+        self.px, self.py, self.pz = px, py, pz
+        self.row_size = py * pz                 # kz split degree
+        self.q = decomp.ry * pz + decomp.rz     # my position in the row
 
         tdtype = {np.dtype("float64"): torch.float64,
                   np.dtype("float32"): torch.float32,
@@ -206,33 +221,62 @@ class PencilDFT(BaseDFT):
 
         from pystella_amd.decomp import get_size_start
         Nx, Ny, Nz = self.grid_shape
+        R = self.row_size
         self.NKz = Nz // 2 + 1 if self.is_real else Nz
         self.rank_shape, _ = decomp.get_rank_shape_start(self.grid_shape)
-        nx_loc, ny_loc, _ = self.rank_shape
+        nx_loc, ny_loc, nz_loc = self.rank_shape
 
-        # splits used by the transposes
-        self.kz_chunks = [get_size_start(self.NKz, py, r)[0]
-                          for r in range(py)]
+        # -- split tables -------------------------------------------------
+        # z-join phase (within the pz-group sharing (rx, ry)): my local
+        # y extent splits into pz chunks; peers' z extents join to Nz
+        self.yz_chunks = [get_size_start(ny_loc, pz, r)[0]
+                          for r in range(pz)]
+        self.z_chunks = [get_size_start(Nz, pz, r)[0] for r in range(pz)]
+        self.ny_z = self.yz_chunks[decomp.rz]   # my y extent after z-join
+
+        # row phase: kz splits into R = py*pz chunks; the row peers' post-
+        # z-join y extents join to Ny (peer position q = ry*pz + rz)
+        self.kz_chunks = [get_size_start(self.NKz, R, r)[0]
+                          for r in range(R)]
+        self.yrow_chunks = []
+        for ry in range(py):
+            ny_r = get_size_start(Ny, py, ry)[0]
+            for rz in range(pz):
+                self.yrow_chunks.append(get_size_start(ny_r, pz, rz)[0])
+        self.kz_loc = self.kz_chunks[self.q]
+
+        # col phase: y splits into px chunks; the col peers' x extents
+        # join to Nx
         self.y2_chunks = [get_size_start(Ny, px, r)[0] for r in range(px)]
         self.x_chunks = [get_size_start(Nx, px, r)[0] for r in range(px)]
-        self.y_chunks = [get_size_start(Ny, py, r)[0] for r in range(py)]
-        self.kz_loc = self.kz_chunks[decomp.ry]
         self.ny2_loc = self.y2_chunks[decomp.rx]
 
-        # communicator subgroups: one per x-row (py-group, fixed rx) and
-        # one per y-column (px-group, fixed ry)
-        self.row_group = None   # ranks sharing rx (vary ry)
-        self.col_group = None   # ranks sharing ry (vary rx)
-        for rx in range(px):
-            ranks = [decomp.rankID(rx, ry, 0) for ry in range(py)]
-            g = dist.new_group(ranks=ranks) if py > 1 else None
-            if rx == decomp.rx:
-                self.row_group = g
-        for ry in range(py):
-            ranks = [decomp.rankID(rx, ry, 0) for rx in range(px)]
-            g = dist.new_group(ranks=ranks) if px > 1 else None
-            if ry == decomp.ry:
-                self.col_group = g
+        # -- communicator subgroups ---------------------------------------
+        # every rank must create every group, in the same order
+        self.z_group = None     # ranks sharing (rx, ry), vary rz
+        self.row_group = None   # ranks sharing rx, vary (ry, rz)
+        self.col_group = None   # ranks sharing (ry, rz), vary rx
+        if pz > 1:
+            for rx in range(px):
+                for ry in range(py):
+                    ranks = [decomp.rankID(rx, ry, rz) for rz in range(pz)]
+                    g = dist.new_group(ranks=ranks)
+                    if rx == decomp.rx and ry == decomp.ry:
+                        self.z_group = g
+        if R > 1:
+            for rx in range(px):
+                ranks = [decomp.rankID(rx, ry, rz)
+                         for ry in range(py) for rz in range(pz)]
+                g = dist.new_group(ranks=ranks)
+                if rx == decomp.rx:
+                    self.row_group = g
+        if px > 1:
+            for ry in range(py):
+                for rz in range(pz):
+                    ranks = [decomp.rankID(rx, ry, rz) for rx in range(px)]
+                    g = dist.new_group(ranks=ranks)
+                    if ry == decomp.ry and rz == decomp.rz:
+                        self.col_group = g
 
         self.fx = torch.empty(self.rank_shape, dtype=tdtype,
                               device=self.device)
@@ -243,7 +287,7 @@ class PencilDFT(BaseDFT):
         ky = fftfreq(Ny)
         kz = rfftfreq(Nz) if self.is_real else fftfreq(Nz)
         _, y2_start = get_size_start(Ny, px, decomp.rx)
-        _, kz_start = get_size_start(self.NKz, py, decomp.ry)
+        _, kz_start = get_size_start(self.NKz, R, self.q)
         self.sub_k = {
             "momenta_x": torch.as_tensor(kx, device=self.device),
             "momenta_y": torch.as_tensor(
@@ -264,42 +308,58 @@ class PencilDFT(BaseDFT):
     # -- transpose helpers --------------------------------------------------
     def _all_to_all(self, chunks_in, group):
         """Exchange a list of tensors (one per peer) within ``group``;
-        returns the received list (same shapes as peers sent)."""
+        returns the received list (shapes from ``self._recv_shapes``).
+        Handles real and complex chunks (complex moves as interleaved
+        re/im pairs — RCCL has no complex dtype)."""
         import torch.distributed as dist
         if group is None:
             return chunks_in
-        flat_in = [torch.view_as_real(c.contiguous()).reshape(-1)
-                   for c in chunks_in]
+        is_c = chunks_in[0].is_complex()
+        w = 2 if is_c else 1
+        if is_c:
+            flat_in = [torch.view_as_real(c.contiguous()).reshape(-1)
+                       for c in chunks_in]
+        else:
+            flat_in = [c.contiguous().reshape(-1) for c in chunks_in]
         in_sizes = [int(c.numel()) for c in flat_in]
         send = torch.cat(flat_in)
-        # peers send chunks whose shapes we can compute; caller passes
-        # via self._recv_shapes set before the call
-        out_sizes = [int(np.prod(s)) * 2 for s in self._recv_shapes]
+        out_sizes = [int(np.prod(s)) * w for s in self._recv_shapes]
         recv = torch.empty(sum(out_sizes), dtype=send.dtype,
                            device=send.device)
         dist.all_to_all_single(recv, send, out_sizes, in_sizes, group=group)
         out = []
         off = 0
         for s in self._recv_shapes:
-            n = int(np.prod(s)) * 2
-            piece = recv[off:off + n].view(*s, 2)
-            out.append(torch.view_as_complex(piece))
+            n = int(np.prod(s)) * w
+            if is_c:
+                piece = recv[off:off + n].view(*s, 2)
+                out.append(torch.view_as_complex(piece))
+            else:
+                out.append(recv[off:off + n].view(*s))
             off += n
         return out
 
     def forward_transform(self, fx, fk):
-        nx_loc, ny_loc, Nz = self.rank_shape[0], self.rank_shape[1], \
-            self.grid_shape[2]
+        nx_loc, ny_loc, nz_loc = self.rank_shape
+        Nz = self.grid_shape[2]
+        t = fx
+        # 0) z-join transpose: split y by pz, join z (real or complex)
+        if self.pz > 1:
+            chunks = list(torch.split(t, self.yz_chunks, dim=1))
+            self._recv_shapes = [(nx_loc, self.ny_z, self.z_chunks[r])
+                                 for r in range(self.pz)]
+            recvd = self._all_to_all(chunks, self.z_group)
+            t = torch.cat(recvd, dim=2)                  # (nx, ny_z, Nz)
         # 1) FFT along z (local)
         if self.is_real:
-            t = torch.fft.rfft(fx, norm="backward")      # (nx, ny, NKz)
+            t = torch.fft.rfft(t, norm="backward")       # (nx, ny_z, NKz)
         else:
-            t = torch.fft.fft(fx, norm="backward")
-        # 2) row transpose: kz split over py, y joined
-        if self.py > 1:
+            t = torch.fft.fft(t, norm="backward")
+        # 2) row transpose: kz split over py*pz, y joined
+        if self.row_size > 1:
             chunks = list(torch.split(t, self.kz_chunks, dim=2))
-            self._recv_shapes = [(nx_loc, self.y_chunks[r], self.kz_loc)
-                                 for r in range(self.py)]
+            self._recv_shapes = [(nx_loc, self.yrow_chunks[r], self.kz_loc)
+                                 for r in range(self.row_size)]
             recvd = self._all_to_all(chunks, self.row_group)
             t = torch.cat(recvd, dim=1)                  # (nx, Ny, kz_loc)
         t = torch.fft.fft(t, dim=1, norm="backward")
@@ -315,27 +375,35 @@ class PencilDFT(BaseDFT):
         return fk
 
     def backward_transform(self, fk, fx):
+        nx_loc, ny_loc, nz_loc = self.rank_shape
         t = torch.fft.ifft(fk, dim=0, norm="forward")
         if self.px > 1:
             chunks = list(torch.split(t, self.x_chunks, dim=0))
             self._recv_shapes = [
-                (self.rank_shape[0], self.y2_chunks[r], self.kz_loc)
+                (nx_loc, self.y2_chunks[r], self.kz_loc)
                 for r in range(self.px)]
             recvd = self._all_to_all(chunks, self.col_group)
             t = torch.cat(recvd, dim=1)                  # (nx, Ny, kz_loc)
         t = torch.fft.ifft(t, dim=1, norm="forward")
-        if self.py > 1:
-            chunks = list(torch.split(t, self.y_chunks, dim=1))
+        if self.row_size > 1:
+            chunks = list(torch.split(t, self.yrow_chunks, dim=1))
             self._recv_shapes = [
-                (self.rank_shape[0], self.rank_shape[1], self.kz_chunks[r])
-                for r in range(self.py)]
+                (nx_loc, self.ny_z, self.kz_chunks[r])
+                for r in range(self.row_size)]
             recvd = self._all_to_all(chunks, self.row_group)
-            t = torch.cat(recvd, dim=2)                  # (nx, ny, NKz)
+            t = torch.cat(recvd, dim=2)                  # (nx, ny_z, NKz)
         if self.is_real:
-            out = torch.fft.irfft(t, n=self.grid_shape[2], norm="forward")
+            t = torch.fft.irfft(t, n=self.grid_shape[2], norm="forward")
         else:
-            out = torch.fft.ifft(t, norm="forward")
-        fx.copy_(out)
+            t = torch.fft.ifft(t, norm="forward")
+        # inverse z-join: split z by pz, join y
+        if self.pz > 1:
+            chunks = list(torch.split(t, self.z_chunks, dim=2))
+            self._recv_shapes = [(nx_loc, self.yz_chunks[r], nz_loc)
+                                 for r in range(self.pz)]
+            recvd = self._all_to_all(chunks, self.z_group)
+            t = torch.cat(recvd, dim=1)                  # (nx, ny, nz)
+        fx.copy_(t)
         return fx
 
 

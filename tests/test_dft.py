@@ -53,15 +53,30 @@ def _pencil_worker(rank, world_size, proc_shape, grid_shape):
 
     fk = fft.dft(fx).clone()
     expect_full = np.fft.rfftn(full)
-    # my k-layout: kx full, ky split over px (rx), kz split over py (ry)
+    # k-layout: kx full, ky split over px (rx), kz split over the
+    # py*pz row dimension at position q = ry*pz + rz
     from pystella_amd.decomp import get_size_start
+    px, py, pz = proc_shape
     Ny = grid_shape[1]
     NKz = grid_shape[2] // 2 + 1
-    ny2, y0 = get_size_start(Ny, proc_shape[0], decomp.rx)
-    nkz, z0 = get_size_start(NKz, proc_shape[1], decomp.ry)
+    q = decomp.ry * pz + decomp.rz
+    ny2, y0 = get_size_start(Ny, px, decomp.rx)
+    nkz, z0 = get_size_start(NKz, py * pz, q)
     expect = expect_full[:, y0:y0 + ny2, z0:z0 + nkz]
     assert np.allclose(fk.numpy(), expect, atol=1e-8), \
         f"rank {rank} fk mismatch"
+
+    # sub_k matches the slice actually held
+    kx = np.fft.fftfreq(grid_shape[0], 1 / grid_shape[0])
+    if grid_shape[0] % 2 == 0:
+        kx[grid_shape[0] // 2] = abs(kx[grid_shape[0] // 2])
+    ky = np.fft.fftfreq(Ny, 1 / Ny)
+    if Ny % 2 == 0:
+        ky[Ny // 2] = abs(ky[Ny // 2])
+    kz = np.fft.rfftfreq(grid_shape[2], 1 / grid_shape[2])
+    assert np.allclose(fft.sub_k["momenta_x"].numpy(), kx)
+    assert np.allclose(fft.sub_k["momenta_y"].numpy(), ky[y0:y0 + ny2])
+    assert np.allclose(fft.sub_k["momenta_z"].numpy(), kz[z0:z0 + nkz])
 
     out = torch.empty(rank_shape, dtype=torch.float64)
     fft.idft(fk, out)
@@ -77,5 +92,56 @@ def test_pencil_fft_slab_y():
     run_distributed(_pencil_worker, 2, args=((1, 2, 1), (8, 8, 8)))
 
 
+def test_pencil_fft_slab_z():
+    run_distributed(_pencil_worker, 2, args=((1, 1, 2), (8, 8, 8)))
+
+
 def test_pencil_fft_uneven():
     run_distributed(_pencil_worker, 2, args=((2, 1, 1), (10, 6, 8)))
+
+
+def test_pencil_fft_uneven_z():
+    run_distributed(_pencil_worker, 2, args=((1, 1, 2), (8, 10, 6)))
+
+
+def test_pencil_fft_2d_yz():
+    run_distributed(_pencil_worker, 4, args=((1, 2, 2), (8, 8, 8)))
+
+
+def test_pencil_fft_3d_222():
+    """The driver's natural N=8 topology (2,2,2): full 3-D pencil."""
+    run_distributed(_pencil_worker, 8, args=((2, 2, 2), (8, 8, 8)))
+
+
+def test_pencil_fft_3d_222_uneven():
+    run_distributed(_pencil_worker, 8, args=((2, 2, 2), (10, 6, 12)))
+
+
+def _pencil_c2c_worker(rank, world_size, proc_shape, grid_shape):
+    decomp = ps.DomainDecomposition(proc_shape, 0, grid_shape=grid_shape)
+    fft = ps.DFT(decomp, grid_shape=grid_shape, dtype=np.complex128)
+    rank_shape, start = decomp.get_rank_shape_start(grid_shape)
+
+    rng = np.random.default_rng(6)
+    full = rng.random(grid_shape) + 1j * rng.random(grid_shape)
+    sl = tuple(slice(s, s + n) for s, n in zip(start, rank_shape))
+    fx = torch.as_tensor(full[sl]).contiguous()
+
+    fk = fft.dft(fx).clone()
+    expect_full = np.fft.fftn(full)
+    from pystella_amd.decomp import get_size_start
+    px, py, pz = proc_shape
+    q = decomp.ry * pz + decomp.rz
+    ny2, y0 = get_size_start(grid_shape[1], px, decomp.rx)
+    nkz, z0 = get_size_start(grid_shape[2], py * pz, q)
+    assert np.allclose(fk.numpy(),
+                       expect_full[:, y0:y0 + ny2, z0:z0 + nkz],
+                       atol=1e-8)
+    out = torch.empty(rank_shape, dtype=torch.complex128)
+    fft.idft(fk, out)
+    assert np.allclose(out.numpy(), full[sl] * np.prod(grid_shape),
+                       atol=1e-6)
+
+
+def test_pencil_fft_c2c_222():
+    run_distributed(_pencil_c2c_worker, 8, args=((2, 2, 2), (8, 8, 8)))

@@ -57,11 +57,12 @@ def test_parseval(grid_shape=(16, 16, 16), L=10.0):
     assert abs(total - expect) / expect < 1e-10
 
 
-def _dist_spectra_worker(rank, world_size):
-    """Distributed PowerSpectra over the pencil FFT (gloo, 2 ranks)
-    equals the single-rank result."""
+def _dist_spectra_worker(rank, world_size, proc_shape=None):
+    """Distributed PowerSpectra over the pencil FFT (gloo) equals the
+    single-rank result (any proc_shape, incl. pz > 1)."""
     grid_shape, L = (16, 16, 16), 10.0
-    decomp = ps.DomainDecomposition((world_size, 1, 1), 0,
+    proc_shape = proc_shape or (world_size, 1, 1)
+    decomp = ps.DomainDecomposition(proc_shape, 0,
                                     grid_shape=grid_shape)
     fft = ps.DFT(decomp, grid_shape=grid_shape, dtype=np.float64)
     dk = tuple(2 * np.pi / L for _ in range(3))
@@ -85,6 +86,7 @@ def _dist_spectra_worker(rank, world_size):
     d1.halo_shape = (0, 0, 0)
     d1.rank_shape = grid_shape
     d1.grid_shape = grid_shape
+    d1._buf_pool = {}
     fft1 = ps.DFT(d1, grid_shape=grid_shape, dtype=np.float64)
     spec1 = ps.PowerSpectra(d1, fft1, dk, L**3)(
         torch.as_tensor(full.copy()), k_power=3)
@@ -94,3 +96,60 @@ def _dist_spectra_worker(rank, world_size):
 def test_distributed_spectra():
     from tests.conftest import run_distributed
     run_distributed(_dist_spectra_worker, 2)
+
+
+def test_distributed_spectra_z_slab():
+    from tests.conftest import run_distributed
+    run_distributed(_dist_spectra_worker, 2, args=((1, 1, 2),))
+
+
+def test_distributed_spectra_222():
+    """The driver's N=8 topology (2,2,2): spectra through the 3-D
+    pencil FFT equal the single-rank result."""
+    from tests.conftest import run_distributed
+    run_distributed(_dist_spectra_worker, 8, args=((2, 2, 2),))
+
+
+def _dist_gw_worker(rank, world_size, proc_shape):
+    """Full GW observables pipeline (6× FFT → TT projection →
+    bin_power) on a distributed decomposition equals single-rank."""
+    grid_shape, L = (16, 16, 16), 10.0
+    h = 1
+    dx = tuple(L / n for n in grid_shape)
+    dk = tuple(2 * np.pi / L for _ in range(3))
+
+    def build(decomp):
+        fft = ps.DFT(decomp, grid_shape=grid_shape, dtype=np.float64)
+        spectra = ps.PowerSpectra(decomp, fft, dk, L**3)
+        proj = ps.Projector(fft, h, dk, dx)
+        return fft, spectra, proj
+
+    decomp = ps.DomainDecomposition(proc_shape, 0,
+                                    grid_shape=grid_shape)
+    fft, spectra, proj = build(decomp)
+
+    rng = np.random.default_rng(11)
+    full = rng.standard_normal((6,) + grid_shape)
+    _, start = decomp.get_rank_shape_start(grid_shape)
+    sl = (slice(None),) + tuple(
+        slice(s, s + n) for s, n in zip(start, decomp.rank_shape))
+    hij = torch.as_tensor(full[sl].copy())
+    spec = spectra.gw(hij, proj, hubble=1.3)
+
+    d1 = ps.DomainDecomposition.__new__(ps.DomainDecomposition)
+    d1.proc_shape = (1, 1, 1)
+    d1.rank, d1.nranks = 0, 1
+    d1.rx = d1.ry = d1.rz = 0
+    d1.halo_shape = (0, 0, 0)
+    d1.rank_shape = grid_shape
+    d1.grid_shape = grid_shape
+    d1._buf_pool = {}
+    fft1, spectra1, proj1 = build(d1)
+    spec1 = spectra1.gw(torch.as_tensor(full.copy()), proj1, hubble=1.3)
+    assert np.allclose(spec, spec1, rtol=1e-9, atol=1e-300), \
+        (rank, np.max(np.abs(spec - spec1)))
+
+
+def test_distributed_gw_spectra_222():
+    from tests.conftest import run_distributed
+    run_distributed(_dist_gw_worker, 8, args=((2, 2, 2),))
