@@ -1041,6 +1041,7 @@ class _NTLapCodegen(_LapCodegen):
         self.preload = {}       # (name, lin) -> register var
         self.store_ctx = False
         self.state_map = state_map or {}
+        self.sectioned_lapv = False
 
     def scalar_param(self, name, idx=()):
         # state scalars are hoisted into per-thread registers once at
@@ -1056,7 +1057,50 @@ class _NTLapCodegen(_LapCodegen):
             reg = self.preload.get((f.name, lin))
             if reg is not None:
                 return reg
+        if (self.sectioned_lapv and f.is_spatial and not any(f.shift)
+                and len(outer_idx) <= 1 and f.name == self.lap_name):
+            lin = int(outer_idx[0]) if outer_idx else 0
+            return f"lapv_{lin}"
         return super().field_access(f, outer_idx)
+
+
+LAPSTAGE_SEC_TEMPLATE = """{defines}
+{preamble}
+#define NRED {nred}
+#define NF {nf}
+extern "C" __global__ __launch_bounds__(TBZ * TBY, MINW) void {name}(
+    {params})
+{{
+    double acc[NRED];
+    {init}
+    const int k = k0b + blockIdx.x * TBZ + (threadIdx.x % TBZ);
+    const int j = j0b + blockIdx.y * TBY + (threadIdx.x / TBZ);
+    const int i0 = i0b + blockIdx.z * XCHUNK;
+    const int i1 = (i0 + XCHUNK < i1b) ? i0 + XCHUNK : i1b;
+    if (k < k1b && j < j1b) {{
+        const long sx = PSY * PSZ;
+        {wrap_decls}
+        double ring[NF][2 * H + 1];
+        #pragma unroll
+        for (int fld = 0; fld < NF; ++fld) {{
+            const double* fp = {fname} + (long)fld * PVOL
+                               + (long)(j + H) * PSZ + (k + H);
+            #pragma unroll
+            for (int p = 0; p < 2 * H; ++p) {{
+                {ring_init}
+            }}
+        }}
+        for (int i = i0; i < i1; ++i) {{
+            {x_off}
+            {site_body}
+            #pragma unroll
+            for (int fld = 0; fld < NF; ++fld)
+                #pragma unroll
+                for (int p = 0; p < 2 * H; ++p)
+                    ring[fld][p] = ring[fld][p + 1];
+        }}
+    }}
+""" + REDUCTION_TAIL_BOX
 
 
 class JitLapStage:
@@ -1068,7 +1112,19 @@ class JitLapStage:
     nontemporal loads and every store is nontemporal (streaming data
     never pollutes the L2, which stays available for stencil-neighbor
     reuse).  This is the hot loop's only kernel
-    (see fusion.StencilRKStepper)."""
+    (see fusion.StencilRKStepper).
+
+    For large stencil families (the 6-component GW h_ij sector), the
+    flat per-site emission keeps all 6 Laplacians + 12 k-preloads +
+    ring live simultaneously — 256 VGPRs, 2 waves/SIMD, ~4.2 TB/s
+    (measured round 1).  ``section_size`` restructures the per-site
+    body into per-component SECTIONS (ring load → Laplacian → RHS →
+    k update → stores for one component, then the next) separated by
+    ``__builtin_amdgcn_sched_barrier(0)`` so the register allocator
+    sees short, non-overlapping live ranges; occupancy goes up at the
+    price of less in-thread ILP (hidden instead by the extra waves).
+    Set ``PYSTELLA_SECTIONS=<n>`` to force a section size (0 = flat
+    emission)."""
 
     # tile candidates, best-first for large grids; smaller tiles win on
     # small per-rank grids (multi-GPU strong scaling) where the big
@@ -1089,10 +1145,10 @@ class JitLapStage:
                  scalar_names, halo, rank_shape, dx, nf, f_name="f",
                  lap_name="lap_f", name="rk_lapstage", tile=None,
                  nt=True, state_map=None, min_waves=1,
-                 periodic=(False, False, False)):
-        from pystella_amd.derivs import _LAP_COEFS
+                 periodic=(False, False, False), section_size=None):
+        import os as _os
         from pystella_amd.field import (
-            Field, Subscript, iter_exprs, walk_expr)
+            Field, Subscript, Variable, is_number, iter_exprs, walk_expr)
         if tile is None:
             tile = self.pick_tile(rank_shape)
         self.rank_shape = tuple(rank_shape)
@@ -1107,7 +1163,35 @@ class JitLapStage:
         cg = _NTLapCodegen(field_args, halo, rank_shape, f_name, lap_name,
                            state_map=state_map)
 
-        lines = []
+        tmp_items = list((tmp_instructions or {}).items())
+        store_items = list(map_dict.items())
+        tmp_set = set()
+        for lhs, _ in tmp_items:
+            tname = lhs.name if hasattr(lhs, "name") else str(lhs)
+            tmp_set.add(tname)
+            cg.tmp_names.add(tname)
+
+        # ---- section-size selection -----------------------------------
+        env = _os.environ.get("PYSTELLA_SECTIONS")
+        if env is not None and env != "":
+            section_size = int(env)
+        entries_trivial = all(is_number(e) for e, _ in entries)
+        if section_size is None:
+            section_size = 0    # default pending A/B measurement
+
+        def _store_comp(lhs):
+            if isinstance(lhs, Subscript) and lhs.index \
+                    and isinstance(lhs.index[0], (int, np.integer)):
+                return int(lhs.index[0])
+            return None
+
+        comps = [_store_comp(lhs) for lhs, _ in store_items]
+        sectioned = bool(
+            section_size and section_size < nf and entries_trivial
+            and all(c is not None and 0 <= c < nf for c in comps))
+
+        # ---- NT preload discovery --------------------------------------
+        nt_used = set()
         if nt:
             # preload every read component of the unpadded streaming
             # arrays (the RK k-arrays) with a nontemporal load
@@ -1115,7 +1199,6 @@ class JitLapStage:
                         if fa.spatial and not fa.padded
                         and fa.name != lap_name
                         and len(fa.outer_shape) <= 1}
-            used = set()
 
             def visit(x):
                 if isinstance(x, Subscript) and \
@@ -1123,54 +1206,225 @@ class JitLapStage:
                         x.aggregate.name in nt_names and \
                         len(x.index) == 1 and \
                         isinstance(x.index[0], int):
-                    used.add((x.aggregate.name, int(x.index[0])))
+                    nt_used.add((x.aggregate.name, int(x.index[0])))
                 elif isinstance(x, Field) and x.name in nt_names \
                         and not x.shape:
-                    used.add((x.name, 0))
+                    nt_used.add((x.name, 0))
 
-            for e in iter_exprs([list((tmp_instructions or {}).values()),
+            for e in iter_exprs([[r for _, r in tmp_items],
                                  [e for e, _ in entries]]):
                 walk_expr(e, visit)
-            for nm, lin in sorted(used):
-                reg = f"pl_{nm}_{lin}"
-                off = "(((long)i*NY + j)*NZ + k)"
-                if lin:
-                    off = f"({lin}L*UVOL + {off})"
-                lines.append(
-                    f"const double {reg} = "
-                    f"__builtin_nontemporal_load(&{nm}[{off}]);")
-                cg.preload[(nm, lin)] = reg
+            for nm, lin in sorted(nt_used):
+                cg.preload[(nm, lin)] = f"pl_{nm}_{lin}"
 
-        for lhs, rhs in (tmp_instructions or {}).items():
-            tname = lhs.name if hasattr(lhs, "name") else str(lhs)
-            cg.tmp_names.add(tname)
-            lines.append(f"const double {tname} = {cg.emit(rhs)};")
+        def _preload_decl(nm, lin):
+            off = "(((long)i*NY + j)*NZ + k)"
+            if lin:
+                off = f"({lin}L*UVOL + {off})"
+            return (f"const double pl_{nm}_{lin} = "
+                    f"__builtin_nontemporal_load(&{nm}[{off}]);")
+
+        # ---- shared emission helpers -----------------------------------
+        self.periodic = tuple(periodic)
+        (wrap_decls, lap_terms, lapc0, x_off, ring_load,
+         ring_init) = _lap_stencil_pieces(h, dx, periodic)
+
+        def _emit_tmps(items, out):
+            for lhs, rhs in items:
+                tname = lhs.name if hasattr(lhs, "name") else str(lhs)
+                out.append(f"const double {tname} = {cg.emit(rhs)};")
+
+        def _emit_stores(items, out):
+            for lhs, rhs in items:
+                val = cg.emit(rhs)
+                cg.store_ctx = True
+                dst = cg.emit(lhs)
+                cg.store_ctx = False
+                if nt:
+                    out.append(
+                        f"__builtin_nontemporal_store({val}, &{dst});")
+                else:
+                    out.append(f"{dst} = {val};")
+
+        def _fp_decl(lin):
+            return (f"    const double* fp = {f_name} + (long)fld * PVOL"
+                    f" + (long)(j + H) * PSZ + (k + H);")
+
+        def _stencil_block(lin, out):
+            out.append(f"double lapv_{lin};")
+            out.append("{")
+            out.append(f"    const int fld = {lin};")
+            out.append(_fp_decl(lin))
+            out.append("    " + ring_load)
+            out.append("    const double* cp = fp + (long)(i + H) * sx;")
+            out.append("    double la = ring[fld][H] * LAPC0;")
+            for t in lap_terms:
+                out.append("    " + t)
+            out.append(f"    lapv_{lin} = la;")
+            out.append("}")
+
+        def _ring_load_block(lin, out):
+            out.append("{")
+            out.append(f"    const int fld = {lin};")
+            out.append(_fp_decl(lin))
+            out.append("    " + ring_load)
+            out.append("}")
+
         init_lines, combine_cases = [], []
         for nm, sidx in sorted((state_map or {}).items()):
             init_lines.append(f"const double st_{nm} = state[{sidx}];")
         for r, (expr, op) in enumerate(entries):
             init_lines.append(f"acc[{r}] = {_OP_INIT[op]};")
-            comb = _OP_COMBINE[op]
-            val = cg.emit(expr)
-            lines.append(
-                "{ const double a = acc[%d]; const double b = %s; "
-                "acc[%d] = %s; }" % (r, val, r, comb))
-            combine_cases.append(f"(r == {r}) ? {comb} : ")
-        for lhs, rhs in map_dict.items():
-            val = cg.emit(rhs)
-            cg.store_ctx = True
-            dst = cg.emit(lhs)
-            cg.store_ctx = False
-            if nt:
-                lines.append(
-                    f"__builtin_nontemporal_store({val}, &{dst});")
-            else:
-                lines.append(f"{dst} = {val};")
+            combine_cases.append(f"(r == {r}) ? {_OP_COMBINE[op]} : ")
         combine = "".join(combine_cases) + "0.0"
 
-        self.periodic = tuple(periodic)
-        (wrap_decls, lap_terms, lapc0, x_off, ring_load,
-         ring_init) = _lap_stencil_pieces(h, dx, periodic)
+        def _entry_lines(out):
+            for r, (expr, op) in enumerate(entries):
+                val = cg.emit(expr)
+                out.append(
+                    "{ const double a = acc[%d]; const double b = %s; "
+                    "acc[%d] = %s; }" % (r, val, r, _OP_COMBINE[op]))
+
+        def _vars_of(expr):
+            out = set()
+
+            def v(x):
+                if isinstance(x, Variable):
+                    out.add(x.name)
+
+            walk_expr(expr, v)
+            return out
+
+        def _lap_lins(exprs):
+            lins = set()
+
+            def v(x):
+                if isinstance(x, Subscript) \
+                        and isinstance(x.aggregate, Field) \
+                        and x.aggregate.name == lap_name and x.index \
+                        and isinstance(x.index[0], (int, np.integer)):
+                    lins.add(int(x.index[0]))
+                elif isinstance(x, Field) and x.name == lap_name:
+                    lins.add(0)
+
+            for e in exprs:
+                walk_expr(e, v)
+            return lins
+
+        def _nt_in(exprs):
+            found = set()
+
+            def v(x):
+                if isinstance(x, Subscript) \
+                        and isinstance(x.aggregate, Field) \
+                        and len(x.index) == 1 \
+                        and isinstance(x.index[0], int):
+                    key = (x.aggregate.name, int(x.index[0]))
+                    if key in cg.preload:
+                        found.add(key)
+                elif isinstance(x, Field) and not x.shape \
+                        and (x.name, 0) in cg.preload:
+                    found.add((x.name, 0))
+
+            for e in exprs:
+                walk_expr(e, v)
+            return found
+
+        if sectioned:
+            # per-component sections: ring load + Laplacian + tmps +
+            # stores for one component at a time, sched_barrier-fenced
+            cg.sectioned_lapv = True
+            n_sec = (nf + section_size - 1) // section_size
+            sec_stores = [[] for _ in range(n_sec)]
+            for (lhs, rhs), c in zip(store_items, comps):
+                sec_stores[c // section_size].append((lhs, rhs))
+
+            deps = {}
+            for lhs, rhs in tmp_items:
+                tname = lhs.name if hasattr(lhs, "name") else str(lhs)
+                deps[tname] = _vars_of(rhs) & tmp_set
+            need = {t: set() for t in tmp_set}
+            for s in range(n_sec):
+                frontier = set()
+                for _, rhs in sec_stores[s]:
+                    frontier |= _vars_of(rhs) & tmp_set
+                seen = set()
+                while frontier:
+                    t = frontier.pop()
+                    if t in seen:
+                        continue
+                    seen.add(t)
+                    need[t].add(s)
+                    frontier |= deps.get(t, set())
+            pro_tmps = []
+            sec_tmps = [[] for _ in range(n_sec)]
+            for lhs, rhs in tmp_items:
+                tname = lhs.name if hasattr(lhs, "name") else str(lhs)
+                regs = need.get(tname, set())
+                if len(regs) == 1:
+                    sec_tmps[next(iter(regs))].append((lhs, rhs))
+                else:
+                    pro_tmps.append((lhs, rhs))
+
+            region_exprs = [
+                [r for _, r in sec_tmps[s]] + [r for _, r in sec_stores[s]]
+                for s in range(n_sec)]
+            pro_exprs = ([r for _, r in pro_tmps]
+                         + [e for e, _ in entries])
+            used_by = {lin: set() for lin in range(nf)}
+            for s in range(n_sec):
+                for lin in _lap_lins(region_exprs[s]):
+                    used_by.setdefault(lin, set()).add(s)
+            for lin in _lap_lins(pro_exprs):
+                used_by.setdefault(lin, set()).add(-1)
+            lin_place = {}
+            for lin in range(nf):
+                regs = used_by.get(lin, set())
+                if len(regs) == 1 and -1 not in regs:
+                    lin_place[lin] = next(iter(regs))
+                elif regs:
+                    lin_place[lin] = -1
+                else:
+                    lin_place[lin] = None    # ring-load only
+
+            nt_region = {key: set() for key in nt_used}
+            for s in range(n_sec):
+                for key in _nt_in(region_exprs[s]):
+                    nt_region[key].add(s)
+            for key in _nt_in(pro_exprs):
+                nt_region[key].add(-1)
+
+            body = []
+            for lin in range(nf):
+                if lin_place[lin] is None:
+                    _ring_load_block(lin, body)
+                elif lin_place[lin] == -1:
+                    _stencil_block(lin, body)
+            for key in sorted(k for k, v in nt_region.items()
+                              if len(v) != 1 or -1 in v):
+                body.append(_preload_decl(*key))
+            _emit_tmps(pro_tmps, body)
+            _entry_lines(body)
+            for s in range(n_sec):
+                body.append("__builtin_amdgcn_sched_barrier(0);")
+                for lin in range(nf):
+                    if lin_place[lin] == s:
+                        _stencil_block(lin, body)
+                for key in sorted(k for k, v in nt_region.items()
+                                  if v == {s}):
+                    body.append(_preload_decl(*key))
+                _emit_tmps(sec_tmps[s], body)
+                _emit_stores(sec_stores[s], body)
+            body.append("__builtin_amdgcn_sched_barrier(0);")
+            lines = body
+        else:
+            lines = []
+            for nm, lin in sorted(nt_used):
+                lines.append(_preload_decl(nm, lin))
+            _emit_tmps(tmp_items, lines)
+            _entry_lines(lines)
+            _emit_stores(store_items, lines)
+        self.sectioned = sectioned
 
         # pointer params: stencil field first, then every other spatial
         # field referenced by the statements or reducers, then the
@@ -1199,14 +1453,23 @@ class JitLapStage:
         defines += f"#define MINW {min_waves}\n"
         defines += f"#define COMBINE(r, a, b) ({combine})\n"
         defines += f"#define LAPC0 ({lapc0!r})\n"
-        src = LAPSTAGE_TEMPLATE.format(
-            defines=defines, preamble=PREAMBLE, nred=len(entries), nf=nf,
-            name=name, params=params, fname=f_name,
-            init="\n    ".join(init_lines),
-            wrap_decls="\n        ".join(wrap_decls),
-            x_off=x_off, ring_load=ring_load, ring_init=ring_init,
-            lap_terms="\n                ".join(lap_terms),
-            body="\n            ".join(lines))
+        if sectioned:
+            src = LAPSTAGE_SEC_TEMPLATE.format(
+                defines=defines, preamble=PREAMBLE, nred=len(entries),
+                nf=nf, name=name, params=params, fname=f_name,
+                init="\n    ".join(init_lines),
+                wrap_decls="\n        ".join(wrap_decls),
+                x_off=x_off, ring_init=ring_init,
+                site_body="\n            ".join(lines))
+        else:
+            src = LAPSTAGE_TEMPLATE.format(
+                defines=defines, preamble=PREAMBLE, nred=len(entries),
+                nf=nf, name=name, params=params, fname=f_name,
+                init="\n    ".join(init_lines),
+                wrap_decls="\n        ".join(wrap_decls),
+                x_off=x_off, ring_load=ring_load, ring_init=ring_init,
+                lap_terms="\n                ".join(lap_terms),
+                body="\n            ".join(lines))
         self.source = src
         self.scalar_keys = [k for _, k in cg.scalars]
         self.key = ext().jit_compile(src, name)
