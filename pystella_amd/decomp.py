@@ -122,14 +122,21 @@ class DomainDecomposition:
                            else tuple(halo_shape))
 
         dist = _dist()
-        if dist.is_initialized():
+        px, py, pz = self.proc_shape
+        if px * py * pz == 1:
+            # a (1,1,1) decomposition is always LOCAL — self-contained
+            # even inside a multi-rank program (single-rank oracles in
+            # distributed tests, per-rank side computations, coarse
+            # gather-to-one-rank solves)
+            self.rank = 0
+            self.nranks = 1
+        elif dist.is_initialized():
             self.rank = dist.get_rank()
             self.nranks = dist.get_world_size()
         else:
             self.rank = 0
             self.nranks = 1
 
-        px, py, pz = self.proc_shape
         if px * py * pz != self.nranks:
             raise ValueError(
                 f"{proc_shape} is an invalid decomposition for "

@@ -174,6 +174,122 @@ def test_multigrid_solve_fp32():
     assert rel < 0.05, rel
 
 
+def _dist_mg_worker(rank, world_size, proc_shape, use_rbgs=False):
+    """FAS V-cycle on a distributed decomposition produces the SAME
+    iterate as the single-rank solve (same smoother, same cycle) —
+    Jacobi/Newton sweeps are rank-count-independent, so the fields
+    must agree to fp roundoff."""
+    n, h, L = 32, 1, 10.0
+
+    # single-rank oracle, computed identically on every rank (a
+    # (1,1,1) decomposition is local even inside a multi-rank program)
+    decomp1, dx, problems1, f_exact, rho_pad = _poisson_setup(n, h, L)
+    Smoother = NewtonIterator
+    kwargs = dict(fixed_parameters=dict(omega=0.8))
+    if use_rbgs:
+        from pystella_amd.multigrid import RedBlackIterator
+        Smoother = RedBlackIterator
+        kwargs = dict(fixed_parameters=dict(omega=1.0))
+    solver1 = Smoother(decomp1, problems1, halo_shape=h, **kwargs)
+    mg1 = FullApproximationScheme(solver1, halo_shape=h)
+    f1 = torch.zeros_like(f_exact)
+    errs1 = mg1(decomp1, dx0=dx, cycle=v_cycle(6, 12, 2),
+                f=f1, rho=rho_pad)
+
+    # distributed solve of the same problem
+    grid_shape = (n, n, n)
+    decomp = ps.DomainDecomposition(proc_shape, h, grid_shape=grid_shape)
+    rank_shape, start = decomp.get_rank_shape_start(grid_shape)
+    pad_sl = tuple(slice(s, s + m + 2 * h)
+                   for s, m in zip(start, rank_shape))
+    f = torch.zeros(tuple(m + 2 * h for m in rank_shape),
+                    dtype=torch.float64)
+    rho_loc = rho_pad[pad_sl].contiguous()
+    decomp.share_halos(rho_loc)
+
+    solver = Smoother(decomp, problems1, halo_shape=h, **kwargs)
+    mg = FullApproximationScheme(solver, halo_shape=h)
+    errs = mg(decomp, dx0=dx, cycle=v_cycle(6, 12, 2),
+              f=f, rho=rho_loc)
+
+    # error histories agree (they are global reductions)
+    for (l1, e1), (l2, e2) in zip(errs1, errs):
+        assert l1 == l2
+        assert np.allclose(e1["f"], e2["f"], rtol=1e-10), (l1, e1, e2)
+
+    # and the final iterate agrees pointwise with the oracle's slice
+    int_sl = tuple(slice(s + h, s + h + m)
+                   for s, m in zip(start, rank_shape))
+    mine = f[(slice(h, -h),) * 3]
+    want = f1[int_sl]
+    err = (mine - want).abs().max().item()
+    assert err < 1e-11, (rank, err)
+
+
+def test_distributed_mg_2rank():
+    from tests.conftest import run_distributed
+    run_distributed(_dist_mg_worker, 2, args=((2, 1, 1),))
+
+
+def test_distributed_mg_2rank_z():
+    from tests.conftest import run_distributed
+    run_distributed(_dist_mg_worker, 2, args=((1, 1, 2),))
+
+
+def test_distributed_mg_222():
+    """The driver's N=8 topology: 32^3 over (2,2,2) coarsens to
+    4^3+halos per rank at the bottom of the V-cycle — deep coarsening
+    across ranks must still match the single-rank solve."""
+    from tests.conftest import run_distributed
+    run_distributed(_dist_mg_worker, 8, args=((2, 2, 2),))
+
+
+def test_distributed_mg_rbgs_2rank():
+    """Red-black Gauss-Seidel across a rank seam: the global
+    checkerboard parity fix (rb_off from the rank's global start) makes
+    the distributed sweep identical to the single-rank sweep."""
+    from tests.conftest import run_distributed
+    run_distributed(_dist_mg_worker, 2, args=((2, 1, 1), True))
+
+
+def _dist_rbgs_odd_worker(rank, world_size):
+    """Standalone RBGS sweeps over an ODD split (30 -> 15+15): rank 1's
+    global x-start is odd, so its checkerboard color must flip
+    (rb_off=1) to agree with the single-rank sweep."""
+    from pystella_amd.multigrid import RedBlackIterator
+    n, h, L = 30, 1, 10.0
+    decomp1, dx, problems, f_exact, rho_pad = _poisson_setup(n, h, L)
+    s1 = RedBlackIterator(decomp1, problems, halo_shape=h,
+                          fixed_parameters=dict(omega=1.0))
+    f1 = torch.zeros_like(f_exact)
+    s1(decomp1, iterations=4, f=f1, tmp_f=torch.zeros_like(f1),
+       rho=rho_pad, dx=np.array(dx))
+
+    grid_shape = (n, n, n)
+    decomp = ps.DomainDecomposition((2, 1, 1), h, grid_shape=grid_shape)
+    rank_shape, start = decomp.get_rank_shape_start(grid_shape)
+    pad_sl = tuple(slice(s, s + m + 2 * h)
+                   for s, m in zip(start, rank_shape))
+    rho_loc = rho_pad[pad_sl].contiguous()
+    decomp.share_halos(rho_loc)
+    f = torch.zeros(tuple(m + 2 * h for m in rank_shape),
+                    dtype=torch.float64)
+    s = RedBlackIterator(decomp, problems, halo_shape=h,
+                         fixed_parameters=dict(omega=1.0))
+    s(decomp, iterations=4, f=f, tmp_f=torch.zeros_like(f),
+      rho=rho_loc, dx=np.array(dx))
+
+    int_sl = tuple(slice(st + h, st + h + m)
+                   for st, m in zip(start, rank_shape))
+    err = (f[(slice(h, -h),) * 3] - f1[int_sl]).abs().max().item()
+    assert err < 1e-12, (rank, err)
+
+
+def test_distributed_rbgs_odd_split():
+    from tests.conftest import run_distributed
+    run_distributed(_dist_rbgs_odd_worker, 2)
+
+
 def test_rbgs_smoother_and_mg():
     """Red-black Gauss-Seidel smoother: converges at least as fast as
     Jacobi per sweep and drives the FAS solver."""
