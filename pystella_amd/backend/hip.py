@@ -1394,6 +1394,12 @@ class JitLapStage:
             for key in _nt_in(pro_exprs):
                 nt_region[key].add(-1)
 
+            # PYSTELLA_SECTION_MASK: instruction categories allowed to
+            # cross the section fences (sched_barrier mask; e.g. 0x20
+            # lets VMEM reads prefetch across sections while keeping
+            # arithmetic sectioned)
+            mask = int(_os.environ.get("PYSTELLA_SECTION_MASK", "0"), 0)
+            fence = f"__builtin_amdgcn_sched_barrier({mask});"
             body = []
             for lin in range(nf):
                 if lin_place[lin] is None:
@@ -1406,7 +1412,7 @@ class JitLapStage:
             _emit_tmps(pro_tmps, body)
             _entry_lines(body)
             for s in range(n_sec):
-                body.append("__builtin_amdgcn_sched_barrier(0);")
+                body.append(fence)
                 for lin in range(nf):
                     if lin_place[lin] == s:
                         _stencil_block(lin, body)
@@ -1415,7 +1421,7 @@ class JitLapStage:
                     body.append(_preload_decl(*key))
                 _emit_tmps(sec_tmps[s], body)
                 _emit_stores(sec_stores[s], body)
-            body.append("__builtin_amdgcn_sched_barrier(0);")
+            body.append(fence)
             lines = body
         else:
             lines = []

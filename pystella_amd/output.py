@@ -109,6 +109,39 @@ class _H5Store:
         self.f.close()
 
 
+class _MiniH5Store:
+    """HDF5 store over the self-contained writer
+    (:mod:`pystella_amd.hdf5`) — same group/dataset/attribute layout as
+    the h5py path, no libhdf5 needed.  The file is rewritten on every
+    append (datasets are contiguous; time-series outputs are small)."""
+
+    def __init__(self, path):
+        from pystella_amd.hdf5 import File
+        self.f = File(path)
+
+    def set_attrs(self, attrs):
+        for k, v in attrs.items():
+            if isinstance(v, (dict, list, tuple)):
+                import json as _json
+                v = _json.dumps(v, default=str)
+            self.f.attrs[k] = v
+        self.f.flush()
+
+    def append(self, group, name, value):
+        self.f.require_group(group)
+        self.f.append(f"{group}/{name}", np.asarray(value))
+        self.f.flush()
+
+    def read(self, group, name):
+        from pystella_amd.hdf5 import read_file
+        tree = read_file(self.f.filename)
+        return np.asarray(
+            tree["children"][group]["children"][name]["data"])
+
+    def close(self):
+        self.f.close()
+
+
 class OutputFile:
     """Appendable run output with provenance attributes
     (reference output.py:52-181)."""
@@ -144,13 +177,19 @@ class OutputFile:
                 pass
         attrs.update({k: str(v) for k, v in kwargs.items()})
 
-        try:
-            import h5py  # noqa: F401
-            self.store = _H5Store(name + ".h5")
-            self.filename = name + ".h5"
-        except ImportError:
+        backend = os.environ.get("PYSTELLA_OUTPUT", "h5")
+        if backend == "dir":
             self.store = _DirStore(name)
             self.filename = name
+        else:
+            try:
+                import h5py  # noqa: F401
+                self.store = _H5Store(name + ".h5")
+            except ImportError:
+                # self-contained HDF5 writer: same file layout, no
+                # libhdf5 in the image (pystella_amd/hdf5.py)
+                self.store = _MiniH5Store(name + ".h5")
+            self.filename = name + ".h5"
         self.store.set_attrs(attrs)
 
     def output(self, group, **datasets):
