@@ -16,11 +16,57 @@ load.
 
 from __future__ import annotations
 
+import json
 import os
 
+import numpy as np
 import torch
 
 __all__ = ["save_checkpoint", "load_checkpoint"]
+
+
+def _h5_save(path, meta, payload):
+    """Write a checkpoint as an HDF5 file (fields under /fields,
+    metadata as root attrs) via the self-contained writer — the same
+    format family as OutputFile (reference output.py's h5py layout)."""
+    from pystella_amd.hdf5 import File
+    tmp = f"{path}.tmp"
+    with File(tmp) as f:
+        f.attrs["mode"] = meta["mode"]
+        f.attrs["proc_shape"] = np.asarray(meta["proc_shape"],
+                                           dtype=np.int64)
+        if meta.get("grid_shape"):
+            f.attrs["grid_shape"] = np.asarray(meta["grid_shape"],
+                                               dtype=np.int64)
+        f.attrs["padded"] = json.dumps(meta["padded"])
+        f.attrs["user_attrs"] = json.dumps(meta["attrs"], default=str)
+        for name, t in payload.items():
+            f.create_dataset(f"fields/{name}", t.numpy())
+    os.replace(tmp, path)
+
+
+def _h5_load(path):
+    """Read a checkpoint written by :func:`_h5_save` (h5py if present,
+    else the self-contained reader)."""
+    try:
+        import h5py
+        with h5py.File(path, "r") as f:
+            meta = {"mode": f.attrs["mode"],
+                    "padded": json.loads(f.attrs["padded"]),
+                    "attrs": json.loads(f.attrs["user_attrs"])}
+            arrays = {k: torch.as_tensor(np.asarray(v))
+                      for k, v in f["fields"].items()}
+            return meta, arrays
+    except ImportError:
+        pass
+    from pystella_amd.hdf5 import read_file
+    tree = read_file(path)
+    meta = {"mode": tree["attrs"]["mode"],
+            "padded": json.loads(tree["attrs"]["padded"]),
+            "attrs": json.loads(tree["attrs"]["user_attrs"])}
+    arrays = {k: torch.as_tensor(v["data"].copy())
+              for k, v in tree["children"]["fields"]["children"].items()}
+    return meta, arrays
 
 
 def _is_padded(t, decomp):
@@ -37,14 +83,18 @@ def save_checkpoint(path, decomp, arrays, attrs=None, mode="gather"):
             "proc_shape": decomp.proc_shape,
             "grid_shape": decomp.grid_shape,
             "padded": {}}
+    h5 = str(path).endswith(".h5")
     if mode == "shard":
         payload = {}
         for name, t in arrays.items():
             padded = _is_padded(t, decomp)
             meta["padded"][name] = padded
             payload[name] = t.cpu()
-        torch.save({"meta": meta, "arrays": payload},
-                   f"{path}.rank{decomp.rank}.pt")
+        if h5:
+            _h5_save(f"{path}.rank{decomp.rank}", meta, payload)
+        else:
+            torch.save({"meta": meta, "arrays": payload},
+                       f"{path}.rank{decomp.rank}.pt")
         decomp.barrier()
         return
 
@@ -57,32 +107,43 @@ def save_checkpoint(path, decomp, arrays, attrs=None, mode="gather"):
         if decomp.rank == 0:
             payload[name] = full.cpu()
     if decomp.rank == 0:
-        tmp = f"{path}.tmp"
-        torch.save({"meta": meta, "arrays": payload}, tmp)
-        os.replace(tmp, path)
+        if h5:
+            _h5_save(path, meta, payload)
+        else:
+            tmp = f"{path}.tmp"
+            torch.save({"meta": meta, "arrays": payload}, tmp)
+            os.replace(tmp, path)
     decomp.barrier()
 
 
 def load_checkpoint(path, decomp, arrays):
     """Restore ``arrays`` (dict name → preallocated tensor) in place."""
-    shard_path = f"{path}.rank{decomp.rank}.pt"
+    h5 = str(path).endswith(".h5")
+    shard_path = (f"{path}.rank{decomp.rank}" if h5
+                  else f"{path}.rank{decomp.rank}.pt")
     if os.path.exists(shard_path):
-        blob = torch.load(shard_path, weights_only=False)
+        if h5:
+            smeta, sarrays = _h5_load(shard_path)
+        else:
+            blob = torch.load(shard_path, weights_only=False)
+            smeta, sarrays = blob["meta"], blob["arrays"]
         for name, t in arrays.items():
-            t.copy_(blob["arrays"][name].to(t.device))
-            if blob["meta"]["padded"].get(name):
+            t.copy_(sarrays[name].to(t.device))
+            if smeta["padded"].get(name):
                 decomp.share_halos(t)
-        return blob["meta"]["attrs"]
+        return smeta["attrs"]
 
-    blob = None
+    garrays = None
     meta = None
     if decomp.rank == 0:
-        blob = torch.load(path, weights_only=False)
-        meta = blob["meta"]
+        if h5:
+            meta, garrays = _h5_load(path)
+        else:
+            blob = torch.load(path, weights_only=False)
+            meta, garrays = blob["meta"], blob["arrays"]
     meta = decomp.bcast(meta, root=0)
     for name, t in arrays.items():
-        full = blob["arrays"][name].to(t.device) if decomp.rank == 0 \
-            else None
+        full = garrays[name].to(t.device) if decomp.rank == 0 else None
         template = full if decomp.rank == 0 else t
         piece = decomp.scatter_array(template)
         if meta["padded"].get(name):

@@ -120,6 +120,58 @@ def test_outputfile_h5_layout(tmp_path):
     assert np.allclose(out.read("energy", "total"), [1.5, 1.4])
 
 
+def test_checkpoint_h5_roundtrip(tmp_path):
+    """Checkpoint into the HDF5 format family (a .h5 path selects the
+    HDF5 writer) and restore, halo-padded arrays included."""
+    import torch
+    import pystella_amd as ps
+    from pystella_amd.checkpoint import save_checkpoint, load_checkpoint
+    h = 1
+    shape = (6, 6, 6)
+    decomp = ps.DomainDecomposition((1, 1, 1), h, rank_shape=shape)
+    pad = tuple(n + 2 * h for n in shape)
+    f = torch.rand((2,) + pad, dtype=torch.float64)
+    decomp.share_halos(f)
+    g = torch.rand(shape, dtype=torch.float64)
+    path = str(tmp_path / "ckpt.h5")
+    save_checkpoint(path, decomp, {"f": f, "g": g},
+                    attrs={"t": 1.25, "step": 12})
+    raw = open(path, "rb").read()
+    assert raw[:8] == b"\x89HDF\r\n\x1a\n"
+
+    f2 = torch.zeros_like(f)
+    g2 = torch.zeros_like(g)
+    attrs = load_checkpoint(path, decomp, {"f": f2, "g": g2})
+    assert attrs["t"] == 1.25 and attrs["step"] == 12
+    assert torch.equal(f2, f)
+    assert torch.equal(g2, g)
+
+
+def _ckpt_h5_worker(rank, world_size, mode, tmpdir):
+    import torch
+    import pystella_amd as ps
+    from pystella_amd.checkpoint import save_checkpoint, load_checkpoint
+    h = 1
+    grid = (8, 8, 8)
+    decomp = ps.DomainDecomposition((2, 1, 1), h, grid_shape=grid)
+    pad = tuple(n + 2 * h for n in decomp.rank_shape)
+    torch.manual_seed(100 + rank)
+    f = torch.rand((2,) + pad, dtype=torch.float64)
+    decomp.share_halos(f)
+    path = os.path.join(tmpdir, f"ck_{mode}.h5")
+    save_checkpoint(path, decomp, {"f": f}, attrs={"t": 0.5}, mode=mode)
+    f2 = torch.zeros_like(f)
+    attrs = load_checkpoint(path, decomp, {"f": f2})
+    assert attrs["t"] == 0.5
+    assert torch.equal(f2, f), (rank, (f2 - f).abs().max())
+
+
+@pytest.mark.parametrize("mode", ["gather", "shard"])
+def test_checkpoint_h5_distributed(tmp_path, mode):
+    from tests.conftest import run_distributed
+    run_distributed(_ckpt_h5_worker, 2, args=(mode, str(tmp_path)))
+
+
 def test_h5py_parity_if_available(tmp_path):
     """If h5py is ever present, the file must open with it and show
     identical structure (the judge can run this off-image)."""
