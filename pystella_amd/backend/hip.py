@@ -663,6 +663,204 @@ def spectra_bin(fk, wbase, bidx, num_bins):
 
 
 # ---------------------------------------------------------------------------
+# fused k-space projector kernels: one launch per operation, ε-basis
+# built in registers (reference pystella/fourier/projectors.py:108-236
+# runs each of these as ONE generated kernel too; round 1 shipped them
+# as multi-launch torch chains — these close K10)
+
+PROJECTOR_PREAMBLE = """
+struct cplx { double x; double y; };
+__device__ inline cplx cmul(cplx a, cplx b)
+{ return {a.x*b.x - a.y*b.y, a.x*b.y + a.y*b.x}; }
+__device__ inline cplx cadd(cplx a, cplx b)
+{ return {a.x + b.x, a.y + b.y}; }
+__device__ inline cplx conjg(cplx a) { return {a.x, -a.y}; }
+__device__ inline cplx cscale(double s, cplx a)
+{ return {s * a.x, s * a.y}; }
+#define LDC(p, c) cplx{ (p)[2*((long)(c)*VOLK + idx)], \\
+                        (p)[2*((long)(c)*VOLK + idx) + 1] }
+#define STC(p, c, v) { (p)[2*((long)(c)*VOLK + idx)] = (v).x; \\
+                       (p)[2*((long)(c)*VOLK + idx) + 1] = (v).y; }
+"""
+
+PROJECTOR_KERNEL = """
+extern "C" __global__ __launch_bounds__(256) void {name}(
+    {params},
+    const double* __restrict__ eff_x,
+    const double* __restrict__ eff_y,
+    const double* __restrict__ eff_z)
+{{
+    const long idx = (long)blockIdx.x * 256 + threadIdx.x;
+    if (idx >= VOLK) return;
+    const int kk = (int)(idx % NKZ);
+    const int jj = (int)((idx / NKZ) % NKY);
+    const int ii = (int)(idx / ((long)NKZ * NKY));
+    const double kx = eff_x[ii], ky = eff_y[jj], kz = eff_z[kk];
+    const double ksq = kx*kx + ky*ky + kz*kz;
+    const bool kzero = (fabs(kx) < 1e-14) && (fabs(ky) < 1e-14)
+                        && (fabs(kz) < 1e-14);
+    (void)ksq; (void)kzero;
+    {eps}
+    {body}
+}}
+"""
+
+# ε-basis construction, matching the torch construction in
+# pystella_amd/fourier/projectors.py:57-78 (reference
+# projectors.py:123-142 incl. the k_x = k_y = 0 special case)
+PROJECTOR_EPS = """
+    const double Kappa = sqrt(kx*kx + ky*ky);
+    const double kmag = sqrt(ksq);
+    const double kmag_s = (kmag > 0.) ? kmag : 1.0;
+    const double Kappa_s = (Kappa > 0.) ? Kappa : 1.0;
+    const bool kxy0 = (fabs(kx) < 1e-10) && (fabs(ky) < 1e-10);
+    const bool kznz = fabs(kz) > 1e-10;
+    const double S2 = 0.7071067811865476;
+    cplx eps0, eps1, eps2;
+    if (kxy0) {
+        eps0 = kznz ? cplx{S2, 0.} : cplx{0., 0.};
+        eps1 = kznz ? cplx{0., S2} : cplx{0., 0.};
+        eps2 = cplx{0., 0.};
+    } else {
+        eps0 = cplx{ kx*kz/kmag_s/Kappa_s*S2, -ky/Kappa_s*S2 };
+        eps1 = cplx{ ky*kz/kmag_s/Kappa_s*S2,  kx/Kappa_s*S2 };
+        eps2 = cplx{ -Kappa/kmag_s*S2, 0. };
+    }
+    (void)eps0; (void)eps1; (void)eps2;
+"""
+
+
+def _proj_sym(c, d):
+    """tid(c,d) for 1-based (c,d) -> symmetric 6-component index."""
+    a, b = min(c, d), max(c, d)
+    return {(1, 1): 0, (1, 2): 1, (1, 3): 2,
+            (2, 2): 3, (2, 3): 4, (3, 3): 5}[(a, b)]
+
+
+def _projector_bodies():
+    eps = ["eps0", "eps1", "eps2"]
+    bodies = {}
+
+    bodies["transversify"] = ("""
+    cplx v0 = LDC(vec,0), v1 = LDC(vec,1), v2 = LDC(vec,2);
+    const double iksq = (ksq > 0.) ? 1.0 / ksq : 0.0;
+    cplx div = cadd(cadd(cscale(kx,v0), cscale(ky,v1)), cscale(kz,v2));
+    cplx r0 = {v0.x - kx*iksq*div.x, v0.y - kx*iksq*div.y};
+    cplx r1 = {v1.x - ky*iksq*div.x, v1.y - ky*iksq*div.y};
+    cplx r2 = {v2.x - kz*iksq*div.x, v2.y - kz*iksq*div.y};
+    if (kzero) { r0 = {0.,0.}; r1 = {0.,0.}; r2 = {0.,0.}; }
+    STC(out,0,r0) STC(out,1,r1) STC(out,2,r2)
+""", ["vec", "out"], False)
+
+    pm_sum = "".join(
+        f"""
+    p = cadd(p, cmul(v{m}, conjg({eps[m]})));
+    mi = cadd(mi, cmul(v{m}, {eps[m]}));""" for m in range(3))
+    bodies["vec_to_pol"] = ("""
+    cplx v0 = LDC(vec,0), v1 = LDC(vec,1), v2 = LDC(vec,2);
+    cplx p = {0.,0.}, mi = {0.,0.};""" + pm_sum + """
+    STC(plus,0,p) STC(minus,0,mi)
+""", ["vec", "plus", "minus"], True)
+
+    p2v = "".join(
+        f"""
+    cplx r{m} = cadd(cmul(p, {eps[m]}), cmul(mi, conjg({eps[m]})));"""
+        for m in range(3))
+    stores = " ".join(f"STC(out,{m},r{m})" for m in range(3))
+    bodies["pol_to_vec"] = ("""
+    cplx p = LDC(plus,0), mi = LDC(minus,0);""" + p2v + f"""
+    {stores}
+""", ["plus", "minus", "out"], True)
+
+    bodies["decompose_vector"] = ("""
+    cplx v0 = LDC(vec,0), v1 = LDC(vec,1), v2 = LDC(vec,2);
+    cplx p = {0.,0.}, mi = {0.,0.};""" + pm_sum + """
+    cplx div = cadd(cadd(cscale(kx,v0), cscale(ky,v1)), cscale(kz,v2));
+    const double denom = (ksq > 0.) ? (TIMES_ABS_K ? sqrt(ksq) : ksq)
+                                     : 1.0;
+    cplx lng = { div.y / denom, -div.x / denom };
+    if (kzero) lng = {0., 0.};
+    STC(plus,0,p) STC(minus,0,mi) STC(lngp,0,lng)
+""", ["vec", "plus", "minus", "lngp"], True)
+
+    d2v = "".join(f"""
+    cplx r{m} = cadd(cmul(p, {eps[m]}), cmul(mi, conjg({eps[m]})));
+    {{ const double km = {"kx" if m == 0 else ("ky" if m == 1 else "kz")};
+       const double fac = TIMES_ABS_K ? km : km / kmag_s;
+       cplx extra = {{ -fac * lng.y, fac * lng.x }};
+       if (!kzero) r{m} = cadd(r{m}, extra); }}""" for m in range(3))
+    bodies["decomp_to_vec"] = ("""
+    cplx p = LDC(plus,0), mi = LDC(minus,0), lng = LDC(lngp,0);"""
+        + d2v + f"""
+    {stores}
+""", ["plus", "minus", "lngp", "out"], True)
+
+    t2p_terms = "".join(
+        f"""
+    p = cadd(p, cmul(h{_proj_sym(c, d)},
+                     cmul(conjg({eps[c-1]}), conjg({eps[d-1]}))));
+    mi = cadd(mi, cmul(h{_proj_sym(c, d)},
+                       cmul({eps[c-1]}, {eps[d-1]})));"""
+        for c in range(1, 4) for d in range(1, 4))
+    loads6 = " ".join(f"cplx h{m} = LDC(hij,{m});" for m in range(6))
+    bodies["tensor_to_pol"] = (f"""
+    {loads6}
+    cplx p = {{0.,0.}}, mi = {{0.,0.}};""" + t2p_terms + """
+    STC(plus,0,p) STC(minus,0,mi)
+""", ["hij", "plus", "minus"], True)
+
+    p2t = "".join(
+        f"""
+    cplx r{_proj_sym(a, b)} = cadd(
+        cmul(p, cmul({eps[a-1]}, {eps[b-1]})),
+        cmul(mi, cmul(conjg({eps[a-1]}), conjg({eps[b-1]}))));"""
+        for a in range(1, 4) for b in range(a, 4))
+    stores6 = " ".join(f"STC(hij,{m},r{m})" for m in range(6))
+    bodies["pol_to_tensor"] = ("""
+    cplx p = LDC(plus,0), mi = LDC(minus,0);""" + p2t + f"""
+    {stores6}
+""", ["plus", "minus", "hij"], True)
+
+    return bodies
+
+
+_PROJ_BODIES = _projector_bodies()
+_proj_cache = {}
+
+
+def projector_source(op, kshape, times_abs_k=False):
+    body, names, needs_eps = _PROJ_BODIES[op]
+    params = ",\n    ".join(f"double* __restrict__ {n}" for n in names)
+    head = (f"#define TIMES_ABS_K {1 if times_abs_k else 0}\n"
+            f"#define NKX {kshape[0]}\n"
+            f"#define NKY {kshape[1]}\n"
+            f"#define NKZ {kshape[2]}\n"
+            "#define VOLK ((long)NKX * NKY * NKZ)\n")
+    return head + PROJECTOR_PREAMBLE + PROJECTOR_KERNEL.format(
+        name=f"proj_{op}", params=params,
+        eps=PROJECTOR_EPS if needs_eps else "", body=body)
+
+
+def projector_op(op, kshape, ptrs, eff, times_abs_k=False):
+    """Launch fused projector kernel ``op`` over k-space ``kshape``.
+
+    :arg ptrs: list of data_ptrs in the op's parameter order.
+    :arg eff: (eff_x, eff_y, eff_z) contiguous fp64 device tensors.
+    """
+    body, names, needs_eps = _PROJ_BODIES[op]
+    key = (op, tuple(kshape), bool(times_abs_k))
+    kid = _proj_cache.get(key)
+    if kid is None:
+        src = projector_source(op, kshape, times_abs_k)
+        kid = ext().jit_compile(src, f"proj_{op}")
+        _proj_cache[key] = kid
+    vol = int(np.prod(kshape))
+    grid = (vol + 255) // 256
+    ext().jit_launch(kid, grid, 1, 1, 256, 1, 1, 0, _stream(),
+                     list(ptrs) + [e.data_ptr() for e in eff], [], [])
+
+
+# ---------------------------------------------------------------------------
 # AOT stencil kernels (csrc/derivs.hip)
 
 def _flat_fields(t, ndim_grid=3):

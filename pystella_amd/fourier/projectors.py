@@ -80,6 +80,30 @@ class Projector:
         khat = tuple(k / kmag_safe for k in self.kvec)
         self.khat = khat
 
+        # contiguous per-axis effective momenta for the fused kernels
+        self._eff_c = tuple(
+            self.eff_mom[n].to(torch.float64).contiguous()
+            for n in ("eff_mom_x", "eff_mom_y", "eff_mom_z"))
+        self._kshape = tuple(len(e) for e in self._eff_c)
+
+    # ------------------------------------------------------------------
+    def _gpu_fast(self, *tensors):
+        """All tensors CUDA complex128 contiguous with this projector's
+        k-space trailing shape → route to the fused one-launch kernels
+        (pystella_amd/backend/hip.py projector_op)."""
+        for t in tensors:
+            if not (isinstance(t, torch.Tensor) and t.is_cuda
+                    and t.dtype == torch.complex128 and t.is_contiguous()
+                    and tuple(t.shape[-3:]) == self._kshape):
+                return False
+        return True
+
+    def _run_op(self, op, tensors, times_abs_k=False):
+        from pystella_amd.backend.hip import projector_op
+        projector_op(op, self._kshape,
+                     [t.data_ptr() for t in tensors], self._eff_c,
+                     times_abs_k=times_abs_k)
+
     # ------------------------------------------------------------------
     def _zero_where_kvec_zero(self, x):
         return torch.where(self.kvec_zero, torch.zeros_like(x), x)
@@ -90,6 +114,9 @@ class Projector:
             vector, vector_T = queue, vector
             queue = None
         out = vector if vector_T is None else vector_T
+        if self._gpu_fast(vector, out):
+            self._run_op("transversify", [vector, out])
+            return out
         div = sum(self.kvec[mu] * vector[mu] for mu in range(3))
         ksq_safe = torch.where(self.ksq > 0, self.ksq,
                                torch.ones_like(self.ksq))
@@ -103,6 +130,9 @@ class Projector:
         if isinstance(queue, torch.Tensor):
             plus, minus, vector = queue, plus, minus
             queue = None
+        if self._gpu_fast(vector, plus, minus):
+            self._run_op("vec_to_pol", [vector, plus, minus])
+            return plus, minus
         p = sum(vector[mu] * self.eps[mu].conj() for mu in range(3))
         m = sum(vector[mu] * self.eps[mu] for mu in range(3))
         plus.copy_(p)
@@ -112,6 +142,9 @@ class Projector:
     def pol_to_vec(self, queue=None, plus=None, minus=None, vector=None):
         if isinstance(queue, torch.Tensor):
             queue, plus, minus, vector = None, queue, plus, minus
+        if self._gpu_fast(plus, minus, vector):
+            self._run_op("pol_to_vec", [plus, minus, vector])
+            return vector
         res = [plus * self.eps[mu] + minus * self.eps[mu].conj()
                for mu in range(3)]
         for mu in range(3):
@@ -125,6 +158,10 @@ class Projector:
         if isinstance(queue, torch.Tensor):
             queue, vector, plus, minus, lng = \
                 None, queue, vector, plus, minus
+        if self._gpu_fast(vector, plus, minus, lng):
+            self._run_op("decompose_vector", [vector, plus, minus, lng],
+                         times_abs_k=times_abs_k)
+            return plus, minus, lng
         self.vec_to_pol(plus=plus, minus=minus, vector=vector)
         div = sum(self.kvec[mu] * vector[mu] for mu in range(3))
         ksq_safe = torch.where(self.ksq > 0, self.ksq,
@@ -141,6 +178,10 @@ class Projector:
         if isinstance(queue, torch.Tensor):
             queue, plus, minus, lng, vector = \
                 None, queue, plus, minus, lng
+        if self._gpu_fast(plus, minus, lng, vector):
+            self._run_op("decomp_to_vec", [plus, minus, lng, vector],
+                         times_abs_k=times_abs_k)
+            return vector
         kmag_safe = torch.where(self.kmag > 0, self.kmag,
                                 torch.ones_like(self.kmag))
         res = []
@@ -202,6 +243,9 @@ class Projector:
     def tensor_to_pol(self, queue=None, plus=None, minus=None, hij=None):
         if isinstance(queue, torch.Tensor):
             queue, plus, minus, hij = None, queue, plus, minus
+        if self._gpu_fast(hij, plus, minus):
+            self._run_op("tensor_to_pol", [hij, plus, minus])
+            return plus, minus
         p = 0
         m = 0
         for c in range(1, 4):
@@ -216,6 +260,9 @@ class Projector:
     def pol_to_tensor(self, queue=None, plus=None, minus=None, hij=None):
         if isinstance(queue, torch.Tensor):
             queue, plus, minus, hij = None, queue, plus, minus
+        if self._gpu_fast(plus, minus, hij):
+            self._run_op("pol_to_tensor", [plus, minus, hij])
+            return hij
         res = []
         for a in range(1, 4):
             for b in range(a, 4):

@@ -1015,3 +1015,106 @@ def test_device_loop_region_split_gpu(grid_shape=(32, 32, 32)):
     assert torch.equal(f_one[cut], f_spl[cut])
     assert st_one["a"] == st_spl["a"]
     assert st_one["energy"] == st_spl["energy"]
+
+
+@requires_gpu
+@pytest.mark.gpu
+def test_fused_projector_kernels_vs_oracle(grid_shape=(24, 20, 16)):
+    """All fused single-launch projector kernels (transversify,
+    vec<->pol, decompose/decomp_to_vec, tensor<->pol; backend/hip.py
+    projector_op) vs the torch oracle path, including aliased
+    outputs (plus/minus as views of the input array, as PowerSpectra
+    uses them)."""
+    from pystella_amd.fourier import DFT
+    decomp = ps.DomainDecomposition((1, 1, 1), 0, rank_shape=grid_shape)
+    dk = (2 * np.pi / 5,) * 3
+    dx = (5 / grid_shape[0],) * 3
+    fft_g = DFT(decomp, grid_shape=grid_shape, dtype=np.float64,
+                device="cuda")
+    fft_c = DFT(decomp, grid_shape=grid_shape, dtype=np.float64,
+                device="cpu")
+    pg = ps.Projector(fft_g, 1, dk, dx)
+    pc = ps.Projector(fft_c, 1, dk, dx)
+    kshape = tuple(fft_c.shape(True))
+    torch.manual_seed(31)
+
+    def rnd(n):
+        return (torch.randn((n,) + kshape, dtype=torch.float64)
+                + 1j * torch.randn((n,) + kshape, dtype=torch.float64)
+                ).to(torch.complex128)
+
+    def check(name, got, want):
+        err = (got.cpu() - want).abs().max().item()
+        scale = max(want.abs().max().item(), 1.0)
+        assert err < 1e-12 * scale, (name, err, scale)
+
+    # transversify (in place)
+    v0 = rnd(3)
+    vc = v0.clone()
+    pc.transversify(vector=vc)
+    vg = v0.clone().cuda().contiguous()
+    pg.transversify(vector=vg)
+    check("transversify", vg, vc)
+
+    # vec_to_pol / pol_to_vec with separate buffers
+    v0 = rnd(3)
+    pl_c = torch.empty(kshape, dtype=torch.complex128)
+    mi_c = torch.empty_like(pl_c)
+    pc.vec_to_pol(plus=pl_c, minus=mi_c, vector=v0.clone())
+    pl_g = torch.empty(kshape, dtype=torch.complex128,
+                       device="cuda")
+    mi_g = torch.empty_like(pl_g)
+    pg.vec_to_pol(plus=pl_g, minus=mi_g, vector=v0.clone().cuda())
+    check("vec_to_pol+", pl_g, pl_c)
+    check("vec_to_pol-", mi_g, mi_c)
+
+    out_c = torch.empty((3,) + kshape, dtype=torch.complex128)
+    pc.pol_to_vec(plus=pl_c, minus=mi_c, vector=out_c)
+    out_g = torch.empty((3,) + kshape, dtype=torch.complex128,
+                        device="cuda")
+    pg.pol_to_vec(plus=pl_g, minus=mi_g, vector=out_g)
+    check("pol_to_vec", out_g, out_c)
+
+    # decompose_vector with ALIASED outputs (views of the input),
+    # exactly as PowerSpectra.vector_decomposition calls it
+    for tak in (False, True):
+        v0 = rnd(3)
+        vc = v0.clone()
+        pc.decompose_vector(vector=vc, plus=vc[0], minus=vc[1],
+                            lng=vc[2], times_abs_k=tak)
+        vg = v0.clone().cuda().contiguous()
+        pg.decompose_vector(vector=vg, plus=vg[0], minus=vg[1],
+                            lng=vg[2], times_abs_k=tak)
+        check(f"decompose_vector tak={tak}", vg, vc)
+
+    # decomp_to_vec
+    for tak in (False, True):
+        d0 = rnd(3)
+        out_c = torch.empty((3,) + kshape, dtype=torch.complex128)
+        pc.decomp_to_vec(plus=d0[0].clone(), minus=d0[1].clone(),
+                         lng=d0[2].clone(), vector=out_c,
+                         times_abs_k=tak)
+        dg = d0.clone().cuda().contiguous()
+        out_g = torch.empty((3,) + kshape, dtype=torch.complex128,
+                            device="cuda")
+        pg.decomp_to_vec(plus=dg[0], minus=dg[1], lng=dg[2],
+                         vector=out_g, times_abs_k=tak)
+        check(f"decomp_to_vec tak={tak}", out_g, out_c)
+
+    # tensor_to_pol with aliased plus/minus (as gw_polarization does)
+    h0 = rnd(6)
+    hc = h0.clone()
+    pc.tensor_to_pol(plus=hc[0], minus=hc[1], hij=hc)
+    hg = h0.clone().cuda().contiguous()
+    pg.tensor_to_pol(plus=hg[0], minus=hg[1], hij=hg)
+    check("tensor_to_pol", hg[:2], hc[:2])
+
+    # pol_to_tensor
+    d0 = rnd(2)
+    h_c = torch.empty((6,) + kshape, dtype=torch.complex128)
+    pc.pol_to_tensor(plus=d0[0].clone(), minus=d0[1].clone(), hij=h_c)
+    dg = d0.clone().cuda().contiguous()
+    h_g = torch.empty((6,) + kshape, dtype=torch.complex128,
+                      device="cuda")
+    pg.pol_to_tensor(plus=dg[0], minus=dg[1], hij=h_g)
+    check("pol_to_tensor", h_g, h_c)

@@ -52,6 +52,30 @@ def main(n=512):
     ms = timeit(lambda: proj.transverse_traceless(hij))
     gb = hij.numel() * 16 * 2 / 1e9
     print(f"TT projection (MFMA)  {ms:8.3f} ms   {gb/ms:5.2f} TB/s")
+
+    # fused one-launch projector family (round 2)
+    vec = hij[:3].clone().contiguous()
+    gbv = vec.numel() * 16 * 2 / 1e9
+    ms = timeit(lambda: proj.transversify(vector=vec))
+    print(f"transversify (fused)  {ms:8.3f} ms   {gbv/ms:5.2f} TB/s")
+    plus, minus = vec[0], vec[1]
+    ms = timeit(lambda: proj.vec_to_pol(plus=plus, minus=minus,
+                                        vector=vec))
+    gb2 = vec.numel() * 16 / 1e9 + 2 * plus.numel() * 16 / 1e9
+    print(f"vec_to_pol (fused)    {ms:8.3f} ms   {gb2/ms:5.2f} TB/s")
+    ms = timeit(lambda: proj.decompose_vector(
+        vector=vec, plus=vec[0], minus=vec[1], lng=vec[2],
+        times_abs_k=True))
+    print(f"decompose_vec (fused) {ms:8.3f} ms   {gb2/ms:5.2f} TB/s")
+    ms = timeit(lambda: proj.tensor_to_pol(plus=hij[0], minus=hij[1],
+                                           hij=hij))
+    gb3 = hij.numel() * 16 / 1e9 + 2 * plus.numel() * 16 / 1e9
+    print(f"tensor_to_pol (fused) {ms:8.3f} ms   {gb3/ms:5.2f} TB/s")
+    vx = torch.rand((3,) + grid, dtype=torch.float64, device=dev)
+    ms = timeit(lambda: spec.polarization(vx, proj), n=5)
+    print(f"full polarization     {ms:8.3f} ms")
+    ms = timeit(lambda: spec.vector_decomposition(vx, proj), n=5)
+    print(f"full vec_decomp       {ms:8.3f} ms")
     ms = timeit(lambda: gen.init_field(fx), n=5)
     print(f"Rayleigh init_field   {ms:8.3f} ms")
 
