@@ -870,11 +870,24 @@ def _flat_fields(t, ndim_grid=3):
     return t, nf
 
 
+_DERIV_DTYPES = {torch.float64: 0, torch.float32: 1}
+
+
+def _deriv_dtype(name, t):
+    code = _DERIV_DTYPES.get(t.dtype)
+    if code is None:
+        raise TypeError(f"{name}: stencil kernels support fp64/fp32, "
+                        f"got {t.dtype}")
+    return code
+
+
 def derivs(fx, lap=None, pdx=None, pdy=None, pdz=None, grd=None, halo=None,
            dx=None, h=None, stream=True):
     if len(set(halo)) != 1:
         raise NotImplementedError("GPU stencils require isotropic halo")
     _check_tensor("fx", fx)
+    dtype = _deriv_dtype("fx", fx)
+    esize = fx.element_size()
     nxp, nyp, nzp = fx.shape[-3:]
     nx, ny, nz = nxp - 2 * h, nyp - 2 * h, nzp - 2 * h
     _, nf = _flat_fields(fx)
@@ -884,6 +897,8 @@ def derivs(fx, lap=None, pdx=None, pdy=None, pdz=None, grd=None, halo=None,
         if t is None:
             return 0
         _check_tensor("out", t)
+        if t.dtype != fx.dtype:
+            raise TypeError(f"output dtype {t.dtype} != input {fx.dtype}")
         return t.data_ptr()
 
     e = ext()
@@ -891,8 +906,8 @@ def derivs(fx, lap=None, pdx=None, pdy=None, pdz=None, grd=None, halo=None,
     # (per-field component stride 3*uvol) or three standalone arrays
     if grd is not None and isinstance(grd, torch.Tensor):
         _check_tensor("grd", grd)
-        gp = grd.data_ptr()
-        px, py, pz_ = gp, gp + 8 * uvol, gp + 16 * uvol
+        gp = ptr(grd)
+        px, py, pz_ = gp, gp + esize * uvol, gp + 2 * esize * uvol
         g_fstride = 3 * uvol
         want_grad = True
     elif pdx is not None and pdy is not None and pdz is not None:
@@ -906,18 +921,22 @@ def derivs(fx, lap=None, pdx=None, pdy=None, pdz=None, grd=None, halo=None,
 
     if lap is not None or want_grad:
         e.gradlap(fx.data_ptr(), ptr(lap), px, py, pz_, g_fstride,
-                  h, nx, ny, nz, nf, dx[0], dx[1], dx[2], _stream())
+                  h, nx, ny, nz, nf, dx[0], dx[1], dx[2], dtype,
+                  _stream())
         return
     # single-axis derivatives
     for axis, out in enumerate((pdx, pdy, pdz)):
         if out is not None:
             e.pd(fx.data_ptr(), ptr(out), h, axis, 0, nx, ny, nz, nf,
-                 dx[axis], _stream())
+                 dx[axis], dtype, _stream())
 
 
 def divergence(vec, div, halo=None, dx=None, h=None):
     _check_tensor("vec", vec)
     _check_tensor("div", div)
+    dtype = _deriv_dtype("vec", vec)
+    if div.dtype != vec.dtype:
+        raise TypeError(f"div dtype {div.dtype} != vec {vec.dtype}")
     nxp, nyp, nzp = vec.shape[-3:]
     nx, ny, nz = nxp - 2 * h, nyp - 2 * h, nzp - 2 * h
     outer = vec.shape[:-4]
@@ -925,11 +944,11 @@ def divergence(vec, div, halo=None, dx=None, h=None):
     from itertools import product
     for s in product(*[range(n) for n in outer]):
         e.pd(vec[s][0].data_ptr(), div[s].data_ptr(), h, 0, 0,
-             nx, ny, nz, 1, dx[0], _stream())
+             nx, ny, nz, 1, dx[0], dtype, _stream())
         e.pd(vec[s][1].data_ptr(), div[s].data_ptr(), h, 1, 1,
-             nx, ny, nz, 1, dx[1], _stream())
+             nx, ny, nz, 1, dx[1], dtype, _stream())
         e.pd(vec[s][2].data_ptr(), div[s].data_ptr(), h, 2, 1,
-             nx, ny, nz, 1, dx[2], _stream())
+             nx, ny, nz, 1, dx[2], dtype, _stream())
 
 
 # ---------------------------------------------------------------------------

@@ -13,7 +13,10 @@
 //  * fused grad+lap: one pass reads f once and emits all four outputs,
 //    reusing each neighbor load for both the first- and second-derivative
 //    coefficient (the reference fuses the same way: derivs.py:334-337).
-//  * fp64 throughout; the kernels are HBM-bandwidth bound by design.
+//  * dtype-generic: every kernel is templated over the element type
+//    (fp64 and fp32 instantiated), matching the reference's loopy dtype
+//    parameter (reference derivs.py:234 takes any dtype); the kernels
+//    are HBM-bandwidth bound by design either way.
 //
 // Centered-difference coefficients of truncation order 2H
 // (reference derivs.py:127-131, 160-165; standard published tables).
@@ -78,11 +81,11 @@ constexpr int BY = 4;
 // Gradient outputs may live inside a packed (nf, 3, nx, ny, nz) "grd"
 // array: g_fstride is the per-field stride of each gradient component
 // (3*uvol when packed, uvol for standalone component arrays).
-template <int H, bool LAP, bool GRAD>
+template <typename T, int H, bool LAP, bool GRAD>
 __global__ __launch_bounds__(BZ * BY) void gradlap_knl(
-    const double *__restrict__ f, double *__restrict__ lap,
-    double *__restrict__ pdx, double *__restrict__ pdy,
-    double *__restrict__ pdz, int64_t g_fstride, int nx, int ny, int nz,
+    const T *__restrict__ f, T *__restrict__ lap,
+    T *__restrict__ pdx, T *__restrict__ pdy,
+    T *__restrict__ pdz, int64_t g_fstride, int nx, int ny, int nz,
     int nxch, int xchunk,
     double inv_dx, double inv_dy, double inv_dz,
     double inv_dx2, double inv_dy2, double inv_dz2)
@@ -94,6 +97,9 @@ __global__ __launch_bounds__(BZ * BY) void gradlap_knl(
     const int i1 = (i0 + xchunk < nx) ? i0 + xchunk : nx;
     if (k >= nz || j >= ny) return;
 
+    const T ix = (T)inv_dx, iy = (T)inv_dy, iz = (T)inv_dz;
+    const T ix2 = (T)inv_dx2, iy2 = (T)inv_dy2, iz2 = (T)inv_dz2;
+
     const int64_t psz = nz + 2 * H;
     const int64_t psy = ny + 2 * H;
     const int64_t sx = psy * psz;                 // padded x-plane stride
@@ -101,52 +107,52 @@ __global__ __launch_bounds__(BZ * BY) void gradlap_knl(
     const int64_t uvol = (int64_t)nx * ny * nz;
 
     // pointer to (x=-H, y=j, z=k) of this field
-    const double *fp = f + (int64_t)fld * pvol + ((int64_t)(j + H)) * psz
-                       + (k + H);
-    double *outl = LAP ? lap + (int64_t)fld * uvol + (int64_t)j * nz + k
-                       : nullptr;
-    double *outx = GRAD ? pdx + fld * g_fstride + (int64_t)j * nz + k
-                        : nullptr;
-    double *outy = GRAD ? pdy + fld * g_fstride + (int64_t)j * nz + k
-                        : nullptr;
-    double *outz = GRAD ? pdz + fld * g_fstride + (int64_t)j * nz + k
-                        : nullptr;
+    const T *fp = f + (int64_t)fld * pvol + ((int64_t)(j + H)) * psz
+                  + (k + H);
+    T *outl = LAP ? lap + (int64_t)fld * uvol + (int64_t)j * nz + k
+                  : nullptr;
+    T *outx = GRAD ? pdx + fld * g_fstride + (int64_t)j * nz + k
+                   : nullptr;
+    T *outy = GRAD ? pdy + fld * g_fstride + (int64_t)j * nz + k
+                   : nullptr;
+    T *outz = GRAD ? pdz + fld * g_fstride + (int64_t)j * nz + k
+                   : nullptr;
     const int64_t so = (int64_t)ny * nz;          // unpadded x stride
 
     // register ring r[p] holds f at x = i - H + p (center plane p = H)
-    double r[2 * H + 1];
+    T r[2 * H + 1];
 #pragma unroll
     for (int p = 0; p < 2 * H; ++p) r[p] = fp[(int64_t)(i0 + p) * sx];
 
     for (int i = i0; i < i1; ++i) {
         r[2 * H] = fp[(int64_t)(i + 2 * H) * sx];
-        const double c = r[H];
-        const double *cp = fp + (int64_t)(i + H) * sx;  // center plane
+        const T c = r[H];
+        const T *cp = fp + (int64_t)(i + H) * sx;  // center plane
 
-        double lap_acc = 0., gx = 0., gy = 0., gz = 0.;
+        T lap_acc = (T)0, gx = (T)0, gy = (T)0, gz = (T)0;
         if (LAP)
-            lap_acc = FD<H>::l(0) * c * (inv_dx2 + inv_dy2 + inv_dz2);
+            lap_acc = (T)FD<H>::l(0) * c * (ix2 + iy2 + iz2);
 
 #pragma unroll
         for (int s = 1; s <= H; ++s) {
-            const double xm = r[H - s], xp = r[H + s];
-            const double ym = cp[-(int64_t)s * psz], yp = cp[(int64_t)s * psz];
-            const double zm = cp[-s], zp = cp[s];
+            const T xm = r[H - s], xp = r[H + s];
+            const T ym = cp[-(int64_t)s * psz], yp = cp[(int64_t)s * psz];
+            const T zm = cp[-s], zp = cp[s];
             if (LAP)
-                lap_acc += FD<H>::l(s) * ((xp + xm) * inv_dx2
-                                          + (yp + ym) * inv_dy2
-                                          + (zp + zm) * inv_dz2);
+                lap_acc += (T)FD<H>::l(s) * ((xp + xm) * ix2
+                                             + (yp + ym) * iy2
+                                             + (zp + zm) * iz2);
             if (GRAD) {
-                gx += FD<H>::g(s) * (xp - xm);
-                gy += FD<H>::g(s) * (yp - ym);
-                gz += FD<H>::g(s) * (zp - zm);
+                gx += (T)FD<H>::g(s) * (xp - xm);
+                gy += (T)FD<H>::g(s) * (yp - ym);
+                gz += (T)FD<H>::g(s) * (zp - zm);
             }
         }
         if (LAP) outl[(int64_t)i * so] = lap_acc;
         if (GRAD) {
-            outx[(int64_t)i * so] = gx * inv_dx;
-            outy[(int64_t)i * so] = gy * inv_dy;
-            outz[(int64_t)i * so] = gz * inv_dz;
+            outx[(int64_t)i * so] = gx * ix;
+            outy[(int64_t)i * so] = gy * iy;
+            outz[(int64_t)i * so] = gz * iz;
         }
         // rotate the ring
 #pragma unroll
@@ -162,16 +168,16 @@ __global__ __launch_bounds__(BZ * BY) void gradlap_knl(
 constexpr int LBZ = 32;
 constexpr int LBY = 8;
 
-template <int H, bool LAP, bool GRAD>
+template <typename T, int H, bool LAP, bool GRAD>
 __global__ __launch_bounds__(LBZ * LBY) void gradlap_lds_knl(
-    const double *__restrict__ f, double *__restrict__ lap,
-    double *__restrict__ pdx, double *__restrict__ pdy,
-    double *__restrict__ pdz, int64_t g_fstride, int nx, int ny, int nz,
+    const T *__restrict__ f, T *__restrict__ lap,
+    T *__restrict__ pdx, T *__restrict__ pdy,
+    T *__restrict__ pdz, int64_t g_fstride, int nx, int ny, int nz,
     int nxch, int xchunk,
     double inv_dx, double inv_dy, double inv_dz,
     double inv_dx2, double inv_dy2, double inv_dz2)
 {
-    __shared__ double tile[LBY + 2 * H][LBZ + 2 * H];
+    __shared__ T tile[LBY + 2 * H][LBZ + 2 * H];
     const int lz = threadIdx.x % LBZ;
     const int ly = threadIdx.x / LBZ;
     const int k = blockIdx.x * LBZ + lz;
@@ -181,6 +187,9 @@ __global__ __launch_bounds__(LBZ * LBY) void gradlap_lds_knl(
     const int i1 = (i0 + xchunk < nx) ? i0 + xchunk : nx;
     const bool active = (k < nz) && (j < ny);
 
+    const T ix = (T)inv_dx, iy = (T)inv_dy, iz = (T)inv_dz;
+    const T ix2 = (T)inv_dx2, iy2 = (T)inv_dy2, iz2 = (T)inv_dz2;
+
     const int64_t psz = nz + 2 * H;
     const int64_t psy = ny + 2 * H;
     const int64_t sx = psy * psz;
@@ -189,19 +198,19 @@ __global__ __launch_bounds__(LBZ * LBY) void gradlap_lds_knl(
 
     const int jc = j < ny ? j : ny - 1;     // clamped (inactive lanes
     const int kc = k < nz ? k : nz - 1;     // still help staging)
-    const double *fbase = f + (int64_t)fld * pvol;
-    const double *fp = fbase + ((int64_t)(jc + H)) * psz + (kc + H);
-    double *outl = LAP ? lap + (int64_t)fld * uvol + (int64_t)jc * nz + kc
-                       : nullptr;
-    double *outx = GRAD ? pdx + fld * g_fstride + (int64_t)jc * nz + kc
-                        : nullptr;
-    double *outy = GRAD ? pdy + fld * g_fstride + (int64_t)jc * nz + kc
-                        : nullptr;
-    double *outz = GRAD ? pdz + fld * g_fstride + (int64_t)jc * nz + kc
-                        : nullptr;
+    const T *fbase = f + (int64_t)fld * pvol;
+    const T *fp = fbase + ((int64_t)(jc + H)) * psz + (kc + H);
+    T *outl = LAP ? lap + (int64_t)fld * uvol + (int64_t)jc * nz + kc
+                  : nullptr;
+    T *outx = GRAD ? pdx + fld * g_fstride + (int64_t)jc * nz + kc
+                   : nullptr;
+    T *outy = GRAD ? pdy + fld * g_fstride + (int64_t)jc * nz + kc
+                   : nullptr;
+    T *outz = GRAD ? pdz + fld * g_fstride + (int64_t)jc * nz + kc
+                   : nullptr;
     const int64_t so = (int64_t)ny * nz;
 
-    double r[2 * H + 1];
+    T r[2 * H + 1];
 #pragma unroll
     for (int p = 0; p < 2 * H; ++p) r[p] = fp[(int64_t)(i0 + p) * sx];
 
@@ -222,33 +231,32 @@ __global__ __launch_bounds__(LBZ * LBY) void gradlap_lds_knl(
         __syncthreads();
         if (active) {
             const int ty = ly + H, tz = lz + H;
-            const double c = r[H];
-            double lap_acc = 0., gx = 0., gy = 0., gz = 0.;
+            const T c = r[H];
+            T lap_acc = (T)0, gx = (T)0, gy = (T)0, gz = (T)0;
             if (LAP)
-                lap_acc = FD<H>::l(0) * c
-                          * (inv_dx2 + inv_dy2 + inv_dz2);
+                lap_acc = (T)FD<H>::l(0) * c * (ix2 + iy2 + iz2);
 #pragma unroll
             for (int s = 1; s <= H; ++s) {
-                const double xm = r[H - s], xp = r[H + s];
-                const double ym = tile[ty - s][tz];
-                const double yp = tile[ty + s][tz];
-                const double zm = tile[ty][tz - s];
-                const double zp = tile[ty][tz + s];
+                const T xm = r[H - s], xp = r[H + s];
+                const T ym = tile[ty - s][tz];
+                const T yp = tile[ty + s][tz];
+                const T zm = tile[ty][tz - s];
+                const T zp = tile[ty][tz + s];
                 if (LAP)
-                    lap_acc += FD<H>::l(s) * ((xp + xm) * inv_dx2
-                                              + (yp + ym) * inv_dy2
-                                              + (zp + zm) * inv_dz2);
+                    lap_acc += (T)FD<H>::l(s) * ((xp + xm) * ix2
+                                                 + (yp + ym) * iy2
+                                                 + (zp + zm) * iz2);
                 if (GRAD) {
-                    gx += FD<H>::g(s) * (xp - xm);
-                    gy += FD<H>::g(s) * (yp - ym);
-                    gz += FD<H>::g(s) * (zp - zm);
+                    gx += (T)FD<H>::g(s) * (xp - xm);
+                    gy += (T)FD<H>::g(s) * (yp - ym);
+                    gz += (T)FD<H>::g(s) * (zp - zm);
                 }
             }
             if (LAP) outl[(int64_t)i * so] = lap_acc;
             if (GRAD) {
-                outx[(int64_t)i * so] = gx * inv_dx;
-                outy[(int64_t)i * so] = gy * inv_dy;
-                outz[(int64_t)i * so] = gz * inv_dz;
+                outx[(int64_t)i * so] = gx * ix;
+                outy[(int64_t)i * so] = gy * iy;
+                outz[(int64_t)i * so] = gz * iz;
             }
         }
 #pragma unroll
@@ -258,9 +266,9 @@ __global__ __launch_bounds__(LBZ * LBY) void gradlap_lds_knl(
 
 // Single-axis first derivative (optionally accumulating, for divergence).
 // AXIS: 0=x, 1=y, 2=z.
-template <int H, int AXIS, bool ACCUM>
+template <typename T, int H, int AXIS, bool ACCUM>
 __global__ __launch_bounds__(BZ * BY) void pd_knl(
-    const double *__restrict__ f, double *__restrict__ out,
+    const T *__restrict__ f, T *__restrict__ out,
     int nx, int ny, int nz, int nxch, int xchunk, double inv_d)
 {
     const int k = blockIdx.x * BZ + (threadIdx.x % BZ);
@@ -270,6 +278,7 @@ __global__ __launch_bounds__(BZ * BY) void pd_knl(
     const int i1 = (i0 + xchunk < nx) ? i0 + xchunk : nx;
     if (k >= nz || j >= ny) return;
 
+    const T inv = (T)inv_d;
     const int64_t psz = nz + 2 * H;
     const int64_t psy = ny + 2 * H;
     const int64_t sx = psy * psz;
@@ -277,19 +286,19 @@ __global__ __launch_bounds__(BZ * BY) void pd_knl(
     const int64_t uvol = (int64_t)nx * ny * nz;
     const int64_t nstride = (AXIS == 0) ? sx : (AXIS == 1) ? psz : 1;
 
-    const double *fp = f + (int64_t)fld * pvol + ((int64_t)(j + H)) * psz
-                       + (k + H);
-    double *op = out + (int64_t)fld * uvol + (int64_t)j * nz + k;
+    const T *fp = f + (int64_t)fld * pvol + ((int64_t)(j + H)) * psz
+                  + (k + H);
+    T *op = out + (int64_t)fld * uvol + (int64_t)j * nz + k;
     const int64_t so = (int64_t)ny * nz;
 
     for (int i = i0; i < i1; ++i) {
-        const double *cp = fp + (int64_t)(i + H) * sx;
-        double g = 0.;
+        const T *cp = fp + (int64_t)(i + H) * sx;
+        T g = (T)0;
 #pragma unroll
         for (int s = 1; s <= H; ++s)
-            g += FD<H>::g(s) * (cp[(int64_t)s * nstride]
-                                - cp[-(int64_t)s * nstride]);
-        g *= inv_d;
+            g += (T)FD<H>::g(s) * (cp[(int64_t)s * nstride]
+                                   - cp[-(int64_t)s * nstride]);
+        g *= inv;
         if (ACCUM)
             op[(int64_t)i * so] += g;
         else
@@ -312,21 +321,10 @@ inline dim3 tile_grid(int ny, int nz, int nf, int nxch)
     return dim3((nz + BZ - 1) / BZ, (ny + BY - 1) / BY, nf * nxch);
 }
 
-}  // namespace
-
-#define DISPATCH_H(H_, ...)                                              \
-    switch (H_) {                                                        \
-    case 1: { constexpr int H = 1; __VA_ARGS__; break; }                 \
-    case 2: { constexpr int H = 2; __VA_ARGS__; break; }                 \
-    case 3: { constexpr int H = 3; __VA_ARGS__; break; }                 \
-    case 4: { constexpr int H = 4; __VA_ARGS__; break; }                 \
-    default: return 1;                                                   \
-    }
-
-extern "C" int pystella_gradlap(
-    const double *f, double *lap, double *pdx, double *pdy, double *pdz,
-    long long g_fstride, int h, int nx, int ny, int nz, int nf,
-    double dx, double dy, double dz, void *stream_)
+template <typename T>
+int gradlap_t(const void *f, void *lap, void *pdx, void *pdy, void *pdz,
+              long long g_fstride, int h, int nx, int ny, int nz, int nf,
+              double dx, double dy, double dz, void *stream_)
 {
     hipStream_t stream = (hipStream_t)stream_;
     const int xchunk = xchunk_size(nx);
@@ -341,29 +339,32 @@ extern "C" int pystella_gradlap(
     const char *env = getenv("PYSTELLA_LDS");
     const bool use_lds = !(env && atoi(env) == 0);
 
+#define ARGS (const T *)f, (T *)lap, (T *)pdx, (T *)pdy, (T *)pdz,      \
+             (int64_t)g_fstride, nx, ny, nz, nxch, xchunk, ix, iy, iz,  \
+             ix2, iy2, iz2
+#define DISPATCH_H(H_, ...)                                              \
+    switch (H_) {                                                        \
+    case 1: { constexpr int H = 1; __VA_ARGS__; break; }                 \
+    case 2: { constexpr int H = 2; __VA_ARGS__; break; }                 \
+    case 3: { constexpr int H = 3; __VA_ARGS__; break; }                 \
+    case 4: { constexpr int H = 4; __VA_ARGS__; break; }                 \
+    default: return 1;                                                   \
+    }
+
     if (use_lds) {
         const dim3 grid((nz + LBZ - 1) / LBZ, (ny + LBY - 1) / LBY,
                         nf * nxch);
         const dim3 block(LBZ * LBY);
         DISPATCH_H(h, {
             if (do_lap && do_grad)
-                hipLaunchKernelGGL((gradlap_lds_knl<H, true, true>), grid,
-                                   block, 0, stream, f, lap, pdx, pdy,
-                                   pdz, (int64_t)g_fstride, nx, ny, nz,
-                                   nxch, xchunk, ix, iy, iz, ix2, iy2,
-                                   iz2);
+                hipLaunchKernelGGL((gradlap_lds_knl<T, H, true, true>),
+                                   grid, block, 0, stream, ARGS);
             else if (do_lap)
-                hipLaunchKernelGGL((gradlap_lds_knl<H, true, false>), grid,
-                                   block, 0, stream, f, lap, pdx, pdy,
-                                   pdz, (int64_t)g_fstride, nx, ny, nz,
-                                   nxch, xchunk, ix, iy, iz, ix2, iy2,
-                                   iz2);
+                hipLaunchKernelGGL((gradlap_lds_knl<T, H, true, false>),
+                                   grid, block, 0, stream, ARGS);
             else
-                hipLaunchKernelGGL((gradlap_lds_knl<H, false, true>), grid,
-                                   block, 0, stream, f, lap, pdx, pdy,
-                                   pdz, (int64_t)g_fstride, nx, ny, nz,
-                                   nxch, xchunk, ix, iy, iz, ix2, iy2,
-                                   iz2);
+                hipLaunchKernelGGL((gradlap_lds_knl<T, H, false, true>),
+                                   grid, block, 0, stream, ARGS);
         });
         return (int)hipGetLastError();
     }
@@ -372,27 +373,22 @@ extern "C" int pystella_gradlap(
     const dim3 block(BZ * BY);
     DISPATCH_H(h, {
         if (do_lap && do_grad)
-            hipLaunchKernelGGL((gradlap_knl<H, true, true>), grid, block, 0,
-                               stream, f, lap, pdx, pdy, pdz,
-                               (int64_t)g_fstride, nx, ny, nz, nxch,
-                               xchunk, ix, iy, iz, ix2, iy2, iz2);
+            hipLaunchKernelGGL((gradlap_knl<T, H, true, true>), grid,
+                               block, 0, stream, ARGS);
         else if (do_lap)
-            hipLaunchKernelGGL((gradlap_knl<H, true, false>), grid, block, 0,
-                               stream, f, lap, pdx, pdy, pdz,
-                               (int64_t)g_fstride, nx, ny, nz, nxch,
-                               xchunk, ix, iy, iz, ix2, iy2, iz2);
+            hipLaunchKernelGGL((gradlap_knl<T, H, true, false>), grid,
+                               block, 0, stream, ARGS);
         else
-            hipLaunchKernelGGL((gradlap_knl<H, false, true>), grid, block, 0,
-                               stream, f, lap, pdx, pdy, pdz,
-                               (int64_t)g_fstride, nx, ny, nz, nxch,
-                               xchunk, ix, iy, iz, ix2, iy2, iz2);
+            hipLaunchKernelGGL((gradlap_knl<T, H, false, true>), grid,
+                               block, 0, stream, ARGS);
     });
     return (int)hipGetLastError();
+#undef ARGS
 }
 
-extern "C" int pystella_pd(
-    const double *f, double *out, int h, int axis, int accum,
-    int nx, int ny, int nz, int nf, double d, void *stream_)
+template <typename T>
+int pd_t(const void *f, void *out, int h, int axis, int accum,
+         int nx, int ny, int nz, int nf, double d, void *stream_)
 {
     hipStream_t stream = (hipStream_t)stream_;
     const int xchunk = xchunk_size(nx);
@@ -401,21 +397,53 @@ extern "C" int pystella_pd(
     const dim3 block(BZ * BY);
     const double inv = 1. / d;
 
+#define PDARGS (const T *)f, (T *)out, nx, ny, nz, nxch, xchunk, inv
     DISPATCH_H(h, {
         switch (axis * 2 + (accum ? 1 : 0)) {
-        case 0: hipLaunchKernelGGL((pd_knl<H, 0, false>), grid, block, 0,
-                                   stream, f, out, nx, ny, nz, nxch, xchunk, inv); break;
-        case 1: hipLaunchKernelGGL((pd_knl<H, 0, true>), grid, block, 0,
-                                   stream, f, out, nx, ny, nz, nxch, xchunk, inv); break;
-        case 2: hipLaunchKernelGGL((pd_knl<H, 1, false>), grid, block, 0,
-                                   stream, f, out, nx, ny, nz, nxch, xchunk, inv); break;
-        case 3: hipLaunchKernelGGL((pd_knl<H, 1, true>), grid, block, 0,
-                                   stream, f, out, nx, ny, nz, nxch, xchunk, inv); break;
-        case 4: hipLaunchKernelGGL((pd_knl<H, 2, false>), grid, block, 0,
-                                   stream, f, out, nx, ny, nz, nxch, xchunk, inv); break;
-        case 5: hipLaunchKernelGGL((pd_knl<H, 2, true>), grid, block, 0,
-                                   stream, f, out, nx, ny, nz, nxch, xchunk, inv); break;
+        case 0: hipLaunchKernelGGL((pd_knl<T, H, 0, false>), grid, block,
+                                   0, stream, PDARGS); break;
+        case 1: hipLaunchKernelGGL((pd_knl<T, H, 0, true>), grid, block,
+                                   0, stream, PDARGS); break;
+        case 2: hipLaunchKernelGGL((pd_knl<T, H, 1, false>), grid, block,
+                                   0, stream, PDARGS); break;
+        case 3: hipLaunchKernelGGL((pd_knl<T, H, 1, true>), grid, block,
+                                   0, stream, PDARGS); break;
+        case 4: hipLaunchKernelGGL((pd_knl<T, H, 2, false>), grid, block,
+                                   0, stream, PDARGS); break;
+        case 5: hipLaunchKernelGGL((pd_knl<T, H, 2, true>), grid, block,
+                                   0, stream, PDARGS); break;
         }
     });
     return (int)hipGetLastError();
+#undef PDARGS
+}
+
+}  // namespace
+
+// dtype: 0 = fp64, 1 = fp32 (matches backend/hip.py derivs()).
+extern "C" int pystella_gradlap(
+    const void *f, void *lap, void *pdx, void *pdy, void *pdz,
+    long long g_fstride, int h, int nx, int ny, int nz, int nf,
+    double dx, double dy, double dz, int dtype, void *stream_)
+{
+    if (dtype == 0)
+        return gradlap_t<double>(f, lap, pdx, pdy, pdz, g_fstride, h,
+                                 nx, ny, nz, nf, dx, dy, dz, stream_);
+    if (dtype == 1)
+        return gradlap_t<float>(f, lap, pdx, pdy, pdz, g_fstride, h,
+                                nx, ny, nz, nf, dx, dy, dz, stream_);
+    return 2;
+}
+
+extern "C" int pystella_pd(
+    const void *f, void *out, int h, int axis, int accum,
+    int nx, int ny, int nz, int nf, double d, int dtype, void *stream_)
+{
+    if (dtype == 0)
+        return pd_t<double>(f, out, h, axis, accum, nx, ny, nz, nf, d,
+                            stream_);
+    if (dtype == 1)
+        return pd_t<float>(f, out, h, axis, accum, nx, ny, nz, nf, d,
+                           stream_);
+    return 2;
 }

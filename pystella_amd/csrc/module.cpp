@@ -39,15 +39,18 @@ namespace py = pybind11;
     } while (0)
 
 // ---------------------------------------------------------------------------
-// AOT kernels (derivs.hip)
-extern "C" int pystella_gradlap(const double *, double *, double *, double *,
-                                double *, long long, int, int, int, int, int,
-                                double, double, double, void *);
-extern "C" int pystella_pd(const double *, double *, int, int, int, int, int,
-                           int, int, double, void *);
+// AOT kernels (derivs.hip); dtype: 0 = fp64, 1 = fp32
+extern "C" int pystella_gradlap(const void *, void *, void *, void *,
+                                void *, long long, int, int, int, int, int,
+                                double, double, double, int, void *);
+extern "C" int pystella_pd(const void *, void *, int, int, int, int, int,
+                           int, int, double, int, void *);
 
 static void check_knl(int err, const char *what)
 {
+    if (err == 2)
+        throw std::runtime_error(std::string(what) +
+                                 ": unsupported dtype (fp64/fp32 only)");
     if (err != 0)
         throw std::runtime_error(std::string(what) + " launch failed: " +
                                  hipGetErrorString((hipError_t)err));
@@ -56,20 +59,21 @@ static void check_knl(int err, const char *what)
 static void gradlap(uintptr_t f, uintptr_t lap, uintptr_t pdx, uintptr_t pdy,
                     uintptr_t pdz, int64_t g_fstride, int h, int nx, int ny,
                     int nz, int nf, double dx, double dy, double dz,
-                    uintptr_t stream)
+                    int dtype, uintptr_t stream)
 {
-    check_knl(pystella_gradlap((const double *)f, (double *)lap,
-                               (double *)pdx, (double *)pdy, (double *)pdz,
+    check_knl(pystella_gradlap((const void *)f, (void *)lap,
+                               (void *)pdx, (void *)pdy, (void *)pdz,
                                (long long)g_fstride, h, nx, ny, nz, nf,
-                               dx, dy, dz, (void *)stream),
+                               dx, dy, dz, dtype, (void *)stream),
               "gradlap");
 }
 
 static void pd(uintptr_t f, uintptr_t out, int h, int axis, int accum,
-               int nx, int ny, int nz, int nf, double d, uintptr_t stream)
+               int nx, int ny, int nz, int nf, double d, int dtype,
+               uintptr_t stream)
 {
-    check_knl(pystella_pd((const double *)f, (double *)out, h, axis, accum,
-                          nx, ny, nz, nf, d, (void *)stream),
+    check_knl(pystella_pd((const void *)f, (void *)out, h, axis, accum,
+                          nx, ny, nz, nf, d, dtype, (void *)stream),
               "pd");
 }
 

@@ -162,8 +162,12 @@ class Projector:
             self._run_op("decompose_vector", [vector, plus, minus, lng],
                          times_abs_k=times_abs_k)
             return plus, minus, lng
-        self.vec_to_pol(plus=plus, minus=minus, vector=vector)
+        # compute the divergence BEFORE vec_to_pol stores: plus/minus
+        # are routinely views of `vector` itself (the reference passes
+        # vec_k[0:2] as outputs, spectra.py:300-303, and its
+        # one-kernel semantics read every input before writing)
         div = sum(self.kvec[mu] * vector[mu] for mu in range(3))
+        self.vec_to_pol(plus=plus, minus=minus, vector=vector)
         ksq_safe = torch.where(self.ksq > 0, self.ksq,
                                torch.ones_like(self.ksq))
         if times_abs_k:

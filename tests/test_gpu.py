@@ -26,28 +26,35 @@ def _to_dev(*tensors):
 
 @requires_gpu
 @pytest.mark.parametrize("h", [1, 2, 3, 4])
-def test_gradlap_vs_cpu(h, grid_shape=(32, 32, 32)):
+@pytest.mark.parametrize("tdtype", [torch.float64, torch.float32])
+def test_gradlap_vs_cpu(h, tdtype, grid_shape=(32, 32, 32)):
+    """AOT stencil kernels (csrc/derivs.hip, dtype-templated) vs the
+    CPU torch path, fp64 and fp32 (the reference is dtype-generic:
+    derivs.py:234)."""
     decomp = ps.DomainDecomposition((1, 1, 1), h, rank_shape=grid_shape)
     dx = (0.1, 0.11, 0.12)
     derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=grid_shape)
     pad = tuple(n + 2 * h for n in grid_shape)
     torch.manual_seed(0)
-    f = torch.rand((2,) + pad, dtype=torch.float64)
+    f = torch.rand((2,) + pad, dtype=tdtype)
+    tol = 1e-12 if tdtype == torch.float64 else 2e-4
 
-    lap_c = torch.zeros((2,) + grid_shape, dtype=torch.float64)
-    grd_c = torch.zeros((2, 3) + grid_shape, dtype=torch.float64)
+    lap_c = torch.zeros((2,) + grid_shape, dtype=tdtype)
+    grd_c = torch.zeros((2, 3) + grid_shape, dtype=tdtype)
     derivs(fx=f.clone(), lap=lap_c, grd=grd_c)
 
     fg = f.clone().cuda()
-    lap_g = torch.zeros((2,) + grid_shape, dtype=torch.float64,
+    lap_g = torch.zeros((2,) + grid_shape, dtype=tdtype,
                         device="cuda")
-    grd_g = torch.zeros((2, 3) + grid_shape, dtype=torch.float64,
+    grd_g = torch.zeros((2, 3) + grid_shape, dtype=tdtype,
                         device="cuda")
     derivs(fx=fg, lap=lap_g, grd=grd_g)
     torch.cuda.synchronize()
 
-    assert (lap_g.cpu() - lap_c).abs().max().item() < 1e-12
-    assert (grd_g.cpu() - grd_c).abs().max().item() < 1e-12
+    assert (lap_g.cpu() - lap_c).abs().max().item() < tol * (
+        1. + lap_c.abs().max().item())
+    assert (grd_g.cpu() - grd_c).abs().max().item() < tol * (
+        1. + grd_c.abs().max().item())
 
 
 @requires_gpu

@@ -117,3 +117,30 @@ def test_effective_momenta(h):
         expect[np.abs(kk) == n // 2] = 0.
         expect[kk == 0] = 0.
         assert np.allclose(kx, expect)
+
+
+def test_decompose_vector_aliased_outputs():
+    """decompose_vector with plus/minus/lng as views of the INPUT
+    vector (how PowerSpectra.vector_decomposition calls it, matching
+    reference spectra.py:300-314) must equal the separate-buffer
+    result — the one-kernel reference semantics read every input
+    before writing (regression: the torch chain used to compute the
+    longitudinal mode from already-overwritten components)."""
+    fft, proj = setup((12, 10, 8))
+    kshape = tuple(fft.shape(True))
+    rng = np.random.default_rng(17)
+    v0 = torch.as_tensor(rng.standard_normal((3,) + kshape)
+                         + 1j * rng.standard_normal((3,) + kshape))
+    for tak in (False, True):
+        plus = torch.empty(kshape, dtype=torch.complex128)
+        minus = torch.empty_like(plus)
+        lng = torch.empty_like(plus)
+        proj.decompose_vector(vector=v0.clone(), plus=plus, minus=minus,
+                              lng=lng, times_abs_k=tak)
+        va = v0.clone()
+        proj.decompose_vector(vector=va, plus=va[0], minus=va[1],
+                              lng=va[2], times_abs_k=tak)
+        assert torch.allclose(va[0], plus, atol=1e-13)
+        assert torch.allclose(va[1], minus, atol=1e-13)
+        assert torch.allclose(va[2], lng, atol=1e-13), \
+            (va[2] - lng).abs().max()
