@@ -27,10 +27,20 @@ parser.add_argument("--box-dim", "-box", type=float, nargs=3,
 parser.add_argument("--kappa", type=float, default=1 / 10)
 parser.add_argument("--mpl", type=float, default=1)
 parser.add_argument("--mphi", type=float, default=1.20e-6)
-parser.add_argument("--mchi", type=float, default=0.)
-parser.add_argument("--gsq", type=float, default=2.5e-7)
-parser.add_argument("--sigma", type=float, default=0.)
-parser.add_argument("--lambda4", type=float, default=0.)
+# per-scalar couplings: pass one value per coupled scalar chi_i
+# (reference CLI surface: scalar_preheating.py:52-58 nargs="*"; this
+# implementation actually supports multiple chis — the reference
+# declares the lists but hardcodes nscalars=2)
+parser.add_argument("--mchi", type=float, nargs="*", default=0.,
+                    help="the mass(es) of coupled scalars")
+parser.add_argument("--gsq", type=float, nargs="*", default=2.5e-7,
+                    help="the 2-2 coupling of phi to other scalars")
+parser.add_argument("--sigma", type=float, nargs="*", default=0.,
+                    help="the trilinear coupling of phi to other scalars")
+parser.add_argument("--lambda4", type=float, nargs="*", default=0.,
+                    help="the quartic self-coupling of other scalars")
+parser.add_argument("--dtype", default="float64",
+                    choices=["float64", "float32"])
 parser.add_argument("--end-time", "-end-t", type=float, default=20)
 parser.add_argument("--end-scale-factor", "-end-a", type=float, default=20)
 parser.add_argument("--gravitational-waves", "-gws", action="store_true")
@@ -51,9 +61,26 @@ def main(args=None):
     dt = p.kappa * min(dx)
     h = p.halo_shape
 
-    nscalars = 2
-    f0 = [.193 * p.mpl, 0]
-    df0 = [-.142231 * p.mpl, 0]
+    def _coupling_list(v):
+        if isinstance(v, (int, float)):
+            return [float(v)]
+        return [float(x) for x in v] or [0.0]
+
+    mchi, gsq, sigma, lambda4 = map(
+        _coupling_list, (p.mchi, p.gsq, p.sigma, p.lambda4))
+    nchi = max(map(len, (mchi, gsq, sigma, lambda4)))
+    for lst in (mchi, gsq, sigma, lambda4):
+        if len(lst) == 1:
+            lst *= nchi
+        elif len(lst) != nchi:
+            raise SystemExit("coupling lists must have equal lengths")
+
+    nscalars = 1 + nchi
+    f0 = [.193 * p.mpl] + [0] * nchi
+    df0 = [-.142231 * p.mpl] + [0] * nchi
+    np_dtype = np.dtype(p.dtype)
+    torch_dtype = (torch.float64 if np_dtype == np.float64
+                   else torch.float32)
     Stepper = ps.LowStorageRK54
 
     ps.init_distributed()
@@ -67,17 +94,20 @@ def main(args=None):
     rank_shape = decomp.rank_shape
     pad = tuple(n + 2 * h for n in rank_shape)
 
-    fft = ps.DFT(decomp, grid_shape=p.grid_shape, dtype=np.float64,
+    fft = ps.DFT(decomp, grid_shape=p.grid_shape, dtype=np_dtype,
                  device=device)
     derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=rank_shape)
 
     def potential(f):
-        phi, chi = f[0], f[1]
-        unscaled = (p.mphi**2 / 2 * phi**2
-                    + p.mchi**2 / 2 * chi**2
-                    + p.gsq / 2 * phi**2 * chi**2
-                    + p.sigma / 2 * phi * chi**2
-                    + p.lambda4 / 4 * chi**4)
+        phi = f[0]
+        unscaled = p.mphi**2 / 2 * phi**2
+        for i in range(nchi):
+            chi = f[1 + i]
+            unscaled = (unscaled
+                        + mchi[i]**2 / 2 * chi**2
+                        + gsq[i] / 2 * phi**2 * chi**2
+                        + sigma[i] / 2 * phi * chi**2
+                        + lambda4[i] / 4 * chi**4)
         return unscaled / p.mphi**2
 
     scalar_sector = ps.ScalarSector(nscalars, potential=potential)
@@ -108,7 +138,7 @@ def main(args=None):
                                     grid_size=grid_size)
     spectra = ps.PowerSpectra(decomp, fft, dk, volume)
     projector = ps.Projector(fft, h, dk, dx)
-    hist = ps.FieldHistogrammer(decomp, 1000, np.float64,
+    hist = ps.FieldHistogrammer(decomp, 1000, np_dtype,
                                 rank_shape=rank_shape)
 
     a_sq_rho = (3 * p.mpl**2 * ps.Field("hubble", indices=[])**2
@@ -138,7 +168,7 @@ def main(args=None):
             output.a_last_spec = expand.a[0]
             if not p.gravitational_waves:
                 derivs(fx=f, grd=dfdx)
-            tmp = torch.empty(rank_shape, dtype=torch.float64,
+            tmp = torch.empty(rank_shape, dtype=torch_dtype,
                               device=device)
             compute_rho(a=expand.a, hubble=expand.hubble, rho=tmp,
                         f=f, dfdt=dfdt, dfdx=dfdx)
@@ -154,17 +184,17 @@ def main(args=None):
     output.a_last_spec = .1
 
     # field allocation & init
-    f = torch.empty((nscalars,) + pad, dtype=torch.float64, device=device)
+    f = torch.empty((nscalars,) + pad, dtype=torch_dtype, device=device)
     dfdt = torch.empty_like(f)
     dfdx = torch.empty((nscalars, 3) + tuple(rank_shape),
-                       dtype=torch.float64, device=device)
+                       dtype=torch_dtype, device=device)
     lap_f = torch.empty((nscalars,) + tuple(rank_shape),
-                        dtype=torch.float64, device=device)
+                        dtype=torch_dtype, device=device)
     if p.gravitational_waves:
-        hij = torch.zeros((6,) + pad, dtype=torch.float64, device=device)
+        hij = torch.zeros((6,) + pad, dtype=torch_dtype, device=device)
         dhijdt = torch.zeros_like(hij)
         lap_hij = torch.zeros((6,) + tuple(rank_shape),
-                              dtype=torch.float64, device=device)
+                              dtype=torch_dtype, device=device)
     else:
         hij = dhijdt = lap_hij = None
 

@@ -558,6 +558,35 @@ extern "C" __global__ __launch_bounds__(256) void {name}(
 """
 
 
+# Large-bin-count fallback: accumulate straight into global memory
+# (device-scope atomics) when NHIST*NBINS doubles exceed the LDS
+# budget of one workgroup (the reference caps the workgroup and merges
+# through global atomics too: histogram.py:69,114-163).
+HISTOGRAM_GLOBAL_TEMPLATE = """{defines}
+{preamble}
+#define NHIST {nhist}
+#define NBINS {nbins}
+extern "C" __global__ __launch_bounds__(256) void {name}(
+    {params})
+{{
+    double* lh = hist;
+    long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    const long total = (long)NX * NY * NZ;
+    const long stride = (long)gridDim.x * blockDim.x;
+    for (; idx < total; idx += stride) {{
+        const int k = (int)(idx % NZ);
+        const long t = idx / NZ;
+        const int j = (int)(t % NY);
+        const int i = (int)(t / NY);
+        {body}
+    }}
+}}
+"""
+
+# one workgroup may use at most 64 KB LDS on gfx9xx
+_HIST_LDS_DOUBLES = 8192
+
+
 class JitHistogram:
     def __init__(self, pairs, num_bins, field_args, scalar_names, halo,
                  rank_shape, name="hist_map"):
@@ -581,7 +610,10 @@ class JitHistogram:
         dbl_params = ", ".join(f"double {c}" for c, _ in cg.scalars)
         params = ", ".join(x for x in (
             ptr_params, "double* __restrict__ hist", dbl_params) if x)
-        src = HISTOGRAM_TEMPLATE.format(
+        template = (HISTOGRAM_TEMPLATE
+                    if self.nhist * num_bins <= _HIST_LDS_DOUBLES
+                    else HISTOGRAM_GLOBAL_TEMPLATE)
+        src = template.format(
             defines=geometry_defines(halo, rank_shape), preamble=PREAMBLE,
             nhist=self.nhist, nbins=num_bins, name=name, params=params,
             body="\n        ".join(body))

@@ -182,3 +182,27 @@ def test_bench_8rank_cpu(tmp_path):
     d = json.loads(line)
     assert d["config"]["parallelism"] == "decomp3d[2, 2, 2]"
     assert d["value"] > 0
+
+
+def test_scalar_preheating_multi_chi(tmp_path):
+    """Per-scalar coupling lists (--gsq with two values -> three
+    scalars): CLI surface matches reference scalar_preheating.py:52-58
+    nargs="*" and actually evolves the extra scalar."""
+    import scalar_preheating
+    os.chdir(tmp_path)
+    expand, energy = scalar_preheating.main(
+        ["--grid-shape", "12", "12", "12", "--end-time", "0.2",
+         "--device", "cpu", "--no-output",
+         "--gsq", "2.5e-7", "1e-7", "--mchi", "0", "1e-7"])
+    assert np.isfinite(energy["total"])
+
+
+def test_scalar_preheating_fp32(tmp_path):
+    """--dtype float32 runs the whole pipeline (fields, stencils, FFT,
+    histogram) in fp32 (reference dtype-generality parity)."""
+    import scalar_preheating
+    os.chdir(tmp_path)
+    expand, energy = scalar_preheating.main(
+        ["--grid-shape", "12", "12", "12", "--end-time", "0.2",
+         "--device", "cpu", "--no-output", "--dtype", "float32"])
+    assert np.isfinite(energy["total"])

@@ -176,6 +176,31 @@ def test_histogram_vs_cpu(grid_shape=(32, 32, 32)):
 
 
 @requires_gpu
+def test_histogram_large_bins_global_atomics(grid_shape=(24, 24, 24)):
+    """num_bins=4096 with 3 simultaneous histograms exceeds the LDS
+    budget (3*4096 doubles > 64 KB) and must route to the
+    global-atomics fallback, matching the CPU result (reference caps
+    its workgroup + merges via global atomics: histogram.py:69,
+    114-163)."""
+    from pystella_amd.backend.hip import _HIST_LDS_DOUBLES
+    decomp = ps.DomainDecomposition((1, 1, 1), 0, rank_shape=grid_shape)
+    torch.manual_seed(5)
+    f = torch.rand(grid_shape, dtype=torch.float64)
+    F = Field("f", offset=0)
+    num_bins = 4096
+    assert 3 * num_bins > _HIST_LDS_DOUBLES
+    pairs = {"a": (F * num_bins, 1),
+             "b": (F * num_bins, F),
+             "c": ((1 - F) * num_bins, F * F)}
+    mk = lambda: ps.Histogrammer(  # noqa: E731
+        decomp, pairs, num_bins, np.float64, halo_shape=0)
+    out_c = mk()(f=f)
+    out_g = mk()(f=f.cuda())
+    for k in out_c:
+        assert np.allclose(out_c[k], out_g[k], rtol=1e-12, atol=1e-9), k
+
+
+@requires_gpu
 def test_wave_equation_gpu_matches_cpu():
     import os
     import sys
