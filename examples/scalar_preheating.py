@@ -47,6 +47,12 @@ parser.add_argument("--gravitational-waves", "-gws", action="store_true")
 parser.add_argument("--device", default=None)
 parser.add_argument("--outfile", default=None)
 parser.add_argument("--no-output", action="store_true")
+# cross-code validation hooks: dump the initial (f, dfdt) realization
+# to an NPZ, or load one instead of drawing a Rayleigh realization —
+# lets this code and the reference run from IDENTICAL initial data
+# (see tools/reference_crosscheck.py)
+parser.add_argument("--save-init", default=None, metavar="FILE.npz")
+parser.add_argument("--load-init", default=None, metavar="FILE.npz")
 
 
 def main(args=None):
@@ -227,6 +233,22 @@ def main(args=None):
     for i in range(nscalars):
         f[i] += f0[i]
         dfdt[i] += df0[i]
+
+    if p.load_init:
+        data = np.load(p.load_init) if decomp.rank == 0 else None
+        for name, t in (("f", f), ("dfdt", dfdt)):
+            src = (torch.as_tensor(np.ascontiguousarray(data[name]))
+                   .to(device=device, dtype=torch_dtype)
+                   if decomp.rank == 0 else t)
+            piece = decomp.scatter_array(src)
+            decomp.restore_halos(t, piece.to(device))
+            decomp.share_halos(t)
+    if p.save_init:
+        full_f = decomp.gather_array(decomp.remove_halos(f))
+        full_df = decomp.gather_array(decomp.remove_halos(dfdt))
+        if decomp.rank == 0:
+            np.savez(p.save_init, f=full_f.cpu().numpy(),
+                     dfdt=full_df.cpu().numpy())
 
     energy = compute_energy(f, dfdt, lap_f, dfdx, float(expand.a[0]))
     expand = ps.Expansion(energy["total"], Stepper, mpl=p.mpl)
