@@ -161,9 +161,13 @@ def test_bench_8rank_cpu(tmp_path):
     import glob
     import json
     import subprocess
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     cmd = [sys.executable, "-m", "torch.distributed.run",
-           "--master-addr", "127.0.0.1", "--master-port", "29833",
+           "--master-addr", "127.0.0.1", "--master-port", str(port),
            "--nnodes=1", "--nproc-per-node", "8",
            "--redirects", "3", "--log-dir", str(tmp_path / "trlogs"),
            os.path.join(repo, "bench.py"),
