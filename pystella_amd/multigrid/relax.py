@@ -120,6 +120,79 @@ class RelaxationBase:
             errs[k][1] = v[1] ** 0.5
         return errs
 
+    # -- integral-constraint machinery ---------------------------------
+    # (reference relax.py:268-320; the reference's solve_constraint
+    # raises NotImplementedError("constraint solving untested") — this
+    # implementation actually solves the shift)
+
+    def _ensure_constraint_kernels(self):
+        if hasattr(self, "_shifter"):
+            return
+        common = dict(halo_shape=self.halo_shape,
+                      fixed_parameters=self.fixed_parameters)
+        f = Field("f", offset="h")
+        tmp = Field("tmp", offset="h")
+        self._shifter = ElementWiseMap(
+            {tmp: var("scale") * f + var("shift")}, **common)
+        from pystella_amd.field import map_expr
+        avg_reducers = {}
+        for fld, (lhs, rho) in self.lhs_dict.items():
+
+            def rename(x, name=fld.name):
+                # shift-preserving rename: stencil reads f(±s) must
+                # become tmp_f(±s), not the unshifted center value
+                if isinstance(x, Field) and x.name == name:
+                    return x.copy(name="tmp_" + name)
+                return x
+
+            avg_reducers[fld.name] = [(map_expr(lhs, rename) - rho,
+                                       "avg")]
+        self._avg_resid = Reduction(self.decomp, avg_reducers,
+                                    halo_shape=self.halo_shape)
+
+    def eval_constraint(self, queue=None, shifts=None, scales=None,
+                        **kwargs):
+        """⟨L(scale·f + shift) − ρ⟩ per unknown, evaluated on shifted
+        copies in the ``tmp_`` arrays (reference relax.py:280-298)."""
+        if shifts is None and queue is not None:
+            shifts = queue
+        self._ensure_constraint_kernels()
+        for name, shift, scale in zip(self.unknown_names,
+                                      np.atleast_1d(shifts),
+                                      np.atleast_1d(scales)):
+            self._shifter(f=kwargs[name], tmp=kwargs["tmp_" + name],
+                          shift=float(shift), scale=float(scale))
+            self.decomp.share_halos(kwargs["tmp_" + name])
+        f0 = kwargs[self.unknown_names[0]]
+        rank_shape = tuple(n - 2 * hh
+                           for n, hh in zip(f0.shape[-3:], self._h3))
+        self._avg_resid.grid_size = float(
+            np.prod(self.decomp.proc_shape) * np.prod(rank_shape))
+        out = self._avg_resid(**kwargs)
+        return np.array([out[n][0] for n in self.unknown_names])
+
+    def solve_constraint(self, queue=None, **kwargs):
+        """Find per-unknown constant shifts so the volume-averaged
+        residual vanishes (⟨L(f + s) − ρ⟩ = 0) and apply them in
+        place.  Fixes the additive nullspace of pure-Neumann/periodic
+        problems (the reference declares this API but its
+        implementation is disabled, relax.py:301-320)."""
+        from scipy.optimize import root
+        self._ensure_constraint_kernels()
+        n = len(self.unknown_names)
+
+        def F(shifts):
+            return self.eval_constraint(shifts=shifts,
+                                        scales=np.ones(n), **kwargs)
+
+        sol = root(F, np.zeros(n), method="hybr", tol=1e-14)
+        if not sol.success:
+            raise RuntimeError(f"constraint solve failed: {sol.message}")
+        for name, shift in zip(self.unknown_names,
+                               np.atleast_1d(sol.x)):
+            kwargs[name] += float(shift)
+        return np.atleast_1d(sol.x)
+
 
 class JacobiIterator(RelaxationBase):
     """Damped Jacobi: f ← (1−ω) f + ω D⁻¹ (ρ − (L−D) f)

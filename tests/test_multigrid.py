@@ -314,3 +314,45 @@ def test_rbgs_smoother_and_mg():
     final = [e for lvl, e in errs if lvl == 0][-1]["f"]
     initial = [e for lvl, e in errs if lvl == 0][0]["f"]
     assert final[1] < 0.1 * initial[1], (initial, final)
+
+
+def test_constraint_machinery():
+    """Integral-constraint solve (reference relax.py:268-320 declares
+    this API but disables its implementation): for L = lap f - m^2 f,
+    shifting f by s changes the mean residual by -m^2 s, so the solve
+    must zero the volume-averaged residual."""
+    n, h, L = 16, 1, 10.0
+    decomp = ps.DomainDecomposition((1, 1, 1), h,
+                                    rank_shape=(n, n, n))
+    dx = L / n
+    msq = 0.7
+
+    def get_lap(f):
+        from pystella_amd.derivs import _LAP_COEFS, centered_diff
+        return sum(centered_diff(f, _LAP_COEFS[h], direction=mu, order=2)
+                   for mu in range(1, 4)) / var("dx")**2
+
+    f = Field("f", offset="h")
+    rho = Field("rho", offset="h")
+    problems = {f: (get_lap(f) - msq * f, rho)}
+    solver = NewtonIterator(decomp, problems, halo_shape=h,
+                            fixed_parameters=dict(omega=0.8))
+
+    rng = np.random.default_rng(3)
+    pad = (n + 2 * h,) * 3
+    fv = torch.as_tensor(rng.standard_normal(pad))
+    decomp.share_halos(fv)
+    rhov = torch.as_tensor(rng.standard_normal(pad))
+    decomp.share_halos(rhov)
+    kw = dict(f=fv, tmp_f=torch.zeros_like(fv), rho=rhov,
+              dx=np.array(dx))
+
+    avg0 = solver.eval_constraint(shifts=np.zeros(1), scales=np.ones(1),
+                                  **kw)
+    shifts = solver.solve_constraint(**kw)
+    avg1 = solver.eval_constraint(shifts=np.zeros(1), scales=np.ones(1),
+                                  **kw)
+    # analytic check: shift = avg0 / (-(-m^2)) = avg0 / -m^2... the
+    # mean residual is linear in s with slope -m^2
+    assert abs(avg1[0]) < 1e-12 * max(1.0, abs(avg0[0])), (avg0, avg1)
+    assert np.allclose(shifts, avg0 / msq, rtol=1e-8), (shifts, avg0)
