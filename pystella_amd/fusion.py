@@ -748,16 +748,17 @@ class DeviceFriedmannLoop:
             for fld in collect_fields(exprs):
                 if fld.is_padded and any(fld.shift):
                     return (False, False, False)
-        # measured on MI355X: at 512^3 the in-kernel wrap costs ~14
-        # VGPRs (occupancy 5 -> 4 waves/SIMD) and loses 2.3%, but at
-        # 128^3 (launch-bound, the strong-scaling regime) it WINS 12%
-        # (0.437 -> 0.391 ms/step).  Default: on for small per-rank
-        # grids; PYSTELLA_PERIODIC=1/0 forces either way.
+        # measured on MI355X (r01 + r02 strong-scaling proxy,
+        # profiles/r02_strong_proxy.txt): in-kernel periodic reads on
+        # non-decomposed axes win +14 % at 128^3 and +3.3 % at 256^3
+        # per-rank (the N=8/N=64 strong-scaling shapes), are neutral at
+        # 512^3 scalar+GW.  Default: on up to 256^3-per-rank volumes;
+        # PYSTELLA_PERIODIC=1/0 forces either way.
         import os
         force = os.environ.get("PYSTELLA_PERIODIC")
         if force == "0":
             return (False, False, False)
-        if force != "1" and int(np.prod(rank_shape)) > 4_000_000:
+        if force != "1" and int(np.prod(rank_shape)) > 17_000_000:
             return (False, False, False)
         px, py, pz = self.decomp.proc_shape
         return (px == 1, py == 1, pz == 1)

@@ -14,6 +14,7 @@ import numbers
 import numpy as np
 
 from pystella_amd.elementwise import ElementWiseMap
+from pystella_amd.stencil import Stencil
 from pystella_amd.field import Field, diff, fabs, var
 from pystella_amd.reduction import Reduction
 
@@ -57,14 +58,16 @@ class RelaxationBase:
         for f, (lhs, rho) in self.lhs_dict.items():
             tmp = Field("tmp_" + f.name, offset=f.offset)
             step_dict[tmp] = self.step_operator(f, lhs, rho)
-        self.stepper = ElementWiseMap(step_dict, **common)
+        # GPU: LDS-staged stencil form (neighbor reads of the
+        # unknown come from workgroup tiles + x register rings)
+        self.stepper = Stencil(step_dict, **common)
 
         # residual: r_f = rho - L(f)
         residual_dict = {}
         for f, (lhs, rho) in self.lhs_dict.items():
             resid = Field("r_" + f.name, offset="h")
             residual_dict[resid] = rho - lhs
-        self.residual = ElementWiseMap(residual_dict, **common)
+        self.residual = Stencil(residual_dict, **common)
 
         # FAS lhs correction: rho = r + L(f)   (on the coarse level)
         tmp_dict = {}
@@ -74,7 +77,7 @@ class RelaxationBase:
             tmp_dict[t] = lhs
             resid = Field("r_" + f.name, offset="h")
             lhs_corr[rho] = resid + t
-        self.lhs_correction = ElementWiseMap(
+        self.lhs_correction = Stencil(
             lhs_corr, tmp_instructions=tmp_dict, **common)
 
         # residual statistics (L_inf and L2)

@@ -1,18 +1,27 @@
-"""Stencil kernels: elementwise maps whose right-hand sides read
-shifted (neighbor) values of halo-padded fields.
+"""Stencil kernels: maps whose right-hand sides read shifted (neighbor)
+values of halo-padded fields, with workgroup LDS prefetch on the GPU.
 
-Analogue of reference pystella/stencil.py:36-143.  In the reference,
-``Stencil`` and ``StreamingStencil`` are separate loopy code paths (a
-workgroup-tile prefetch kernel vs an x-streaming variant).  Here both
-collapse onto one implementation: the CDNA4 elementwise template
-already walks x with a (64z × 4y) tile per block (the "streaming"
-shape), and shifted reads of padded fields are part of the expression
-language — neighbor loads are served by L1/L2 with coalesced z access.
-``lsize``/``prefetch_args`` keyword arguments are accepted for API
-compatibility and ignored.
+Analogue of reference pystella/stencil.py:36-143.  The reference's
+``Stencil`` prefetches the input bounding box (incl. the ±h ghost rim)
+into workgroup-local memory; ``StreamingStencil`` streams tiles along
+one axis while keeping the cross-section resident.  The CDNA4
+realization here does both at once (backend/hip.py ``JitStencil``):
+blocks march along x (the streaming axis), each neighbor-read field's
+current x-plane tile — including the ghost rim — is staged in LDS per
+iteration, and pure-x-shifted reads are served from a per-thread
+register ring.  Measured 17 % faster than plain L1 neighbor reuse for
+the isolated Laplacian on gfx950 (profiles/r01_lds_vs_ring.txt).
+
+The CPU path and kernels with no prefetchable reads (or whose LDS
+tiles would overflow the budget) fall back to the elementwise form.
+``lsize`` and ``prefetch_args`` keyword arguments are accepted for
+reference API compatibility (the prefetch set is inferred from the
+expressions; explicit tile sizes are fixed per gfx950).
 """
 
 from __future__ import annotations
+
+import torch
 
 from pystella_amd.elementwise import ElementWiseMap
 
@@ -26,6 +35,26 @@ class Stencil(ElementWiseMap):
                          halo_shape=halo_shape, rank_shape=rank_shape,
                          **kwargs)
 
+    def _call_hip(self, env, rank_shape):
+        from pystella_amd.backend.hip import get_stencil_kernel
+        dtype = None
+        for fa in self.field_args:
+            t = env.get(fa.name)
+            if isinstance(t, torch.Tensor) and fa.spatial:
+                dtype = t.dtype
+                break
+        if dtype is None:
+            dtype = torch.float64
+        if self._hip_kernel is None or \
+                self._hip_kernel.rank_shape != rank_shape or \
+                self._hip_kernel.dtype != dtype:
+            self._hip_kernel = get_stencil_kernel(
+                self.map_dict, self.tmp_instructions, self.field_args,
+                sorted(self.scalar_names), self.halo_shape, rank_shape,
+                name=self.name, dtype=dtype)
+        self._hip_kernel(env)
+
 
 class StreamingStencil(Stencil):
-    pass
+    """Same engine: the x-march IS the streaming axis
+    (reference stencil.py:103-141)."""
