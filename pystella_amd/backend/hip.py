@@ -1603,6 +1603,10 @@ class JitLapStage:
 
     @classmethod
     def pick_tile(cls, rank_shape):
+        import os as _os
+        env = _os.environ.get("PYSTELLA_TILE")
+        if env:
+            return tuple(int(x) for x in env.split(","))
         for tile in cls.TILE_CANDIDATES:
             g = _tile_grid(tile, rank_shape)
             if g[0] * g[1] * g[2] >= 2048:
