@@ -36,6 +36,9 @@ class Stencil(ElementWiseMap):
                          **kwargs)
 
     def _call_hip(self, env, rank_shape):
+        import os
+        if os.environ.get("PYSTELLA_STENCIL_LDS") == "0":
+            return super()._call_hip(env, rank_shape)
         from pystella_amd.backend.hip import get_stencil_kernel
         dtype = None
         for fa in self.field_args:
