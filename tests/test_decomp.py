@@ -202,6 +202,17 @@ def test_halo_exchange_2d_pencil():
     run_distributed(_halo_worker, 4, args=((2, 2, 1), (8, 8, 8), 2))
 
 
+def test_overlap_halo_exchange_222():
+    """8 ranks, (2,2,2): ALL axes remote — one batched group carries
+    12 concurrent ops per rank, exactly the pattern the N=8 bench's
+    overlapped path posts over RCCL."""
+    run_distributed(_overlap_halo_worker, 8, args=((2, 2, 2), (8, 8, 8), 2))
+
+
+def test_overlap_halo_exchange_2d():
+    run_distributed(_overlap_halo_worker, 4, args=((2, 2, 1), (8, 8, 8), 1))
+
+
 def test_halo_exchange_3d():
     """8 ranks, full 3-D (2,2,2) decomposition (beyond the reference,
     which caps at 2-D)."""
