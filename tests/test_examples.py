@@ -171,7 +171,7 @@ def test_bench_8rank_cpu(tmp_path):
            "--nnodes=1", "--nproc-per-node", "8",
            "--redirects", "3", "--log-dir", str(tmp_path / "trlogs"),
            os.path.join(repo, "bench.py"),
-           "--gpus", "8", "--steps", "1", "--warmup", "0",
+           "--gpus", "8", "--steps", "2", "--warmup", "1",
            "--grid", "16", "--device", "cpu"]
     env = {k: v for k, v in os.environ.items()
            if k not in ("RANK", "LOCAL_RANK", "WORLD_SIZE",
