@@ -19,6 +19,7 @@ The flat public API mirrors the reference package
 
 from pystella_amd.field import (  # noqa: F401
     Field, DynamicField, index_fields, shift_fields, substitute, diff,
+    collect_field_indices, indices_to_domain, infer_field_domains,
     get_field_args, var,
 )
 from pystella_amd.decomp import DomainDecomposition, init_distributed  # noqa: F401

@@ -30,6 +30,8 @@ __all__ = [
     "Field", "DynamicField", "shift_fields", "substitute", "collect_fields",
     "get_field_args", "FieldArg", "diff", "Expr", "Variable", "Subscript",
     "Sum", "Product", "Quotient", "Power", "Call", "Comparison", "If", "var",
+    "index_fields", "collect_field_indices", "indices_to_domain",
+    "infer_field_domains",
 ]
 
 
@@ -305,6 +307,31 @@ def index_fields(expr, prepend_with=None):
     unchanged (pystella_amd consumes Fields directly; there is no
     subscript-lowering step)."""
     return expr
+
+
+def collect_field_indices(expressions):
+    """All spatial index names appearing in the expressions' Fields
+    (reference field/__init__.py:collect_field_indices; there the
+    result feeds loopy ISL domains — here geometry is compile-time, so
+    this is introspection only)."""
+    indices = set()
+    for f in collect_fields(expressions):
+        indices |= set(f.indices)
+    return tuple(sorted(indices))
+
+
+def indices_to_domain(indices):
+    """ISL-style domain string for the given index names (reference
+    analogue; informational — kernels bake geometry at compile time)."""
+    names = ", ".join(indices)
+    bounds = " and ".join(f"0 <= {i} < N{i}" for i in indices)
+    return f"{{[{names}]: {bounds}}}" if indices else "{[]}"
+
+
+def infer_field_domains(expressions):
+    """Reference-API analogue (field/__init__.py:633): domain string
+    inferred from the expressions' field indices."""
+    return indices_to_domain(collect_field_indices(expressions))
 
 
 from pystella_amd.field.diff import diff  # noqa: E402  (cycle-free)
