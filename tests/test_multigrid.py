@@ -356,3 +356,36 @@ def test_constraint_machinery():
     # mean residual is linear in s with slope -m^2
     assert abs(avg1[0]) < 1e-12 * max(1.0, abs(avg0[0])), (avg0, avg1)
     assert np.allclose(shifts, avg0 / msq, rtol=1e-8), (shifts, avg0)
+
+
+def test_bench_mg_distributed_cpu(tmp_path):
+    """tools/bench_mg.py under torchrun (2 ranks, gloo): BASELINE
+    config 5 names multigrid on 8x MI355X — the distributed launch
+    path must work end to end."""
+    import json
+    import os
+    import subprocess
+    import sys
+    import glob
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [sys.executable, "-m", "torch.distributed.run",
+           "--standalone", "--local-addr", "127.0.0.1",
+           "--nnodes=1", "--nproc-per-node", "2",
+           "--redirects", "3", "--log-dir", str(tmp_path / "trlogs"),
+           os.path.join(repo, "tools", "bench_mg.py"),
+           "--grid", "32", "--cycles", "1", "--depth", "2",
+           "--device", "cpu", "--dtype", "float64",
+           "--smoother", "newton"]
+    env = {k: v for k, v in os.environ.items()
+           if k not in ("RANK", "LOCAL_RANK", "WORLD_SIZE",
+                        "MASTER_ADDR", "MASTER_PORT", "LOCAL_WORLD_SIZE",
+                        "GROUP_RANK", "TORCHELASTIC_RUN_ID")}
+    out = subprocess.run(cmd, capture_output=True, text=True,
+                         timeout=600, cwd=repo, env=env)
+    logs = "\n".join(open(f).read() for f in glob.glob(
+        str(tmp_path / "trlogs" / "**" / "*.log"), recursive=True))
+    assert out.returncode == 0, (out.stderr[-1500:], logs[-1500:])
+    line = [ln for ln in logs.splitlines() if ln.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["n_ranks"] == 2
+    assert d["resid_L2_end"] < d["resid_L2_start"]

@@ -21,7 +21,8 @@ from pystella_amd.multigrid import (  # noqa: E402
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--n", type=int, default=1024)
+    ap.add_argument("--n", "--grid", dest="n", type=int,
+                    default=1024)
     ap.add_argument("--dtype", default="float32")
     ap.add_argument("--cycles", type=int, default=4)
     ap.add_argument("--depth", type=int, default=5)
@@ -36,7 +37,17 @@ def main():
                                 else "cpu"))
     n, h = p.n, 1
     grid = (n, n, n)
-    decomp = ps.DomainDecomposition((1, 1, 1), h, rank_shape=grid)
+    # distributed: launch under torchrun (one rank per GPU); the
+    # decomposition follows WORLD_SIZE like bench.py
+    import os
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world > 1:
+        ps.init_distributed()
+        if device.type == "cuda":
+            device = ps.choose_device()
+    shapes = {1: (1, 1, 1), 2: (2, 1, 1), 4: (2, 2, 1), 8: (2, 2, 2)}
+    proc_shape = shapes.get(world, (world, 1, 1))
+    decomp = ps.DomainDecomposition(proc_shape, h, grid_shape=grid)
     L = 2 * np.pi
     dx = (L / n,) * 3
 
@@ -57,11 +68,14 @@ def main():
     mg = FullApproximationScheme(solver, halo_shape=h)
 
     # manufactured solution: f* = sin(x)sin(y)sin(z), rho = -3 f*
-    ax = torch.arange(n, dtype=torch.float64) * dx[0]
-    s1 = torch.sin(ax)
-    f_exact = (s1[:, None, None] * s1[None, :, None]
-               * s1[None, None, :])
-    pad = (n + 2 * h,) * 3
+    # (rank-local slice of the global analytic fields)
+    rank_shape, start = decomp.get_rank_shape_start(grid)
+    axes = [(torch.arange(s, s + m, dtype=torch.float64) * d)
+            for s, m, d in zip(start, rank_shape, dx)]
+    f_exact = (torch.sin(axes[0])[:, None, None]
+               * torch.sin(axes[1])[None, :, None]
+               * torch.sin(axes[2])[None, None, :])
+    pad = tuple(m + 2 * h for m in rank_shape)
     rho_t = torch.zeros(pad, dtype=torch.float64)
     rho_t[h:-h, h:-h, h:-h] = -3.0 * f_exact
     rho_t = rho_t.to(dtype).to(device)
@@ -76,12 +90,15 @@ def main():
     errs = one_cycle()   # warmup (includes setup + JIT)
     if device.type == "cuda":
         torch.cuda.synchronize()
+    decomp.barrier()
     t0 = time.perf_counter()
     for _ in range(p.cycles):
         errs = one_cycle()
     if device.type == "cuda":
         torch.cuda.synchronize()
+    decomp.barrier()
     dtime = (time.perf_counter() - t0) / p.cycles
+    dtime = float(decomp.allreduce(dtime, op="max"))
 
     final = [e for lvl, e in errs if lvl == 0][-1]["f"]
     initial = [e for lvl, e in errs if lvl == 0][0]["f"]
@@ -90,7 +107,10 @@ def main():
     want = f_exact - f_exact.mean()
     rel = (got - want).abs().max().item() / want.abs().max().item()
     import json
+    if decomp.rank != 0:
+        return
     print(json.dumps({
+        "n_ranks": decomp.nranks,
         "metric": "seconds per FAS V-cycle, Poisson",
         "value": dtime, "unit": "s", "higher_is_better": False,
         "grid": list(grid), "dtype": p.dtype, "depth": p.depth,
