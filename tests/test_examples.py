@@ -210,3 +210,24 @@ def test_scalar_preheating_fp32(tmp_path):
         ["--grid-shape", "12", "12", "12", "--end-time", "0.2",
          "--device", "cpu", "--no-output", "--dtype", "float32"])
     assert np.isfinite(energy["total"])
+
+
+def _preheating_gws_8rank_worker(rank, world_size):
+    """GW example incl. spectra output at the (2,2,2) N=8 topology:
+    the full observables path (pencil FFT pz>1, TT projection, binned
+    spectra, HDF5 output on rank 0) end to end."""
+    import tempfile
+    import scalar_preheating
+    os.chdir(tempfile.mkdtemp())
+    expand, energy = scalar_preheating.main(
+        ["--grid-shape", "16", "16", "16",
+         "--proc-shape", "2", "2", "2",
+         "--end-time", "0.05", "--end-scale-factor", "1.0001",
+         "--device", "cpu", "--gravitational-waves",
+         "--outfile", f"gwout"])
+    assert np.isfinite(energy["total"])
+
+
+def test_scalar_preheating_gws_8rank():
+    from tests.conftest import run_distributed
+    run_distributed(_preheating_gws_8rank_worker, 8)
