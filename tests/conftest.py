@@ -47,3 +47,24 @@ def _dist_worker(rank, world_size, init_file, fn, args):
         fn(rank, world_size, *args)
     finally:
         dist.destroy_process_group()
+
+
+def run_torchrun(cmd, cwd, log_dir, attempts=2, timeout=600):
+    """Run a torchrun command, retrying once on failure (multi-process
+    rendezvous is occasionally flaky under CI load); returns
+    (CompletedProcess, combined log text)."""
+    import glob
+    import subprocess
+    env = {k: v for k, v in __import__("os").environ.items()
+           if k not in ("RANK", "LOCAL_RANK", "WORLD_SIZE",
+                        "MASTER_ADDR", "MASTER_PORT", "LOCAL_WORLD_SIZE",
+                        "GROUP_RANK", "TORCHELASTIC_RUN_ID")}
+    out = None
+    for _ in range(attempts):
+        out = subprocess.run(cmd, capture_output=True, text=True,
+                             timeout=timeout, cwd=cwd, env=env)
+        logs = "\n".join(open(f).read() for f in glob.glob(
+            str(log_dir) + "/**/*.log", recursive=True))
+        if out.returncode == 0:
+            return out, logs
+    return out, logs

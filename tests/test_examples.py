@@ -110,15 +110,8 @@ def test_bench_distributed_cpu(tmp_path):
            os.path.join(repo, "bench.py"),
            "--gpus", "2", "--steps", "2", "--warmup", "1",
            "--grid", "16", "--device", "cpu"]
-    env = {k: v for k, v in os.environ.items()
-           if k not in ("RANK", "LOCAL_RANK", "WORLD_SIZE",
-                        "MASTER_ADDR", "MASTER_PORT", "LOCAL_WORLD_SIZE",
-                        "GROUP_RANK", "TORCHELASTIC_RUN_ID")}
-    out = subprocess.run(cmd, capture_output=True, text=True,
-                         timeout=600, cwd=repo, env=env)
-    import glob
-    logs = "\n".join(open(f).read() for f in glob.glob(
-        str(tmp_path / "trlogs" / "**" / "*.log"), recursive=True))
+    from tests.conftest import run_torchrun
+    out, logs = run_torchrun(cmd, repo, tmp_path / "trlogs")
     assert out.returncode == 0, (out.stderr[-1500:], logs[-1500:])
     line = [ln for ln in logs.splitlines() if ln.startswith("{")][-1]
     d = json.loads(line)
@@ -141,14 +134,8 @@ def test_bench_distributed_gws_cpu(tmp_path):
            os.path.join(repo, "bench.py"),
            "--gpus", "2", "--steps", "1", "--warmup", "0",
            "--grid", "16", "--device", "cpu", "--gws"]
-    env = {k: v for k, v in os.environ.items()
-           if k not in ("RANK", "LOCAL_RANK", "WORLD_SIZE",
-                        "MASTER_ADDR", "MASTER_PORT", "LOCAL_WORLD_SIZE",
-                        "GROUP_RANK", "TORCHELASTIC_RUN_ID")}
-    out = subprocess.run(cmd, capture_output=True, text=True,
-                         timeout=600, cwd=repo, env=env)
-    logs = "\n".join(open(f).read() for f in glob.glob(
-        str(tmp_path / "trlogs" / "**" / "*.log"), recursive=True))
+    from tests.conftest import run_torchrun
+    out, logs = run_torchrun(cmd, repo, tmp_path / "trlogs")
     assert out.returncode == 0, (out.stderr[-1500:], logs[-1500:])
     line = [ln for ln in logs.splitlines() if ln.startswith("{")][-1]
     d = json.loads(line)
@@ -173,14 +160,8 @@ def test_bench_8rank_cpu(tmp_path):
            os.path.join(repo, "bench.py"),
            "--gpus", "8", "--steps", "2", "--warmup", "1",
            "--grid", "16", "--device", "cpu"]
-    env = {k: v for k, v in os.environ.items()
-           if k not in ("RANK", "LOCAL_RANK", "WORLD_SIZE",
-                        "MASTER_ADDR", "MASTER_PORT", "LOCAL_WORLD_SIZE",
-                        "GROUP_RANK", "TORCHELASTIC_RUN_ID")}
-    out = subprocess.run(cmd, capture_output=True, text=True,
-                         timeout=600, cwd=repo, env=env)
-    logs = "\n".join(open(f).read() for f in glob.glob(
-        str(tmp_path / "trlogs" / "**" / "*.log"), recursive=True))
+    from tests.conftest import run_torchrun
+    out, logs = run_torchrun(cmd, repo, tmp_path / "trlogs")
     assert out.returncode == 0, (out.stderr[-1500:], logs[-1500:])
     line = [ln for ln in logs.splitlines() if ln.startswith("{")][-1]
     d = json.loads(line)

@@ -376,14 +376,8 @@ def test_bench_mg_distributed_cpu(tmp_path):
            "--grid", "32", "--cycles", "1", "--depth", "2",
            "--device", "cpu", "--dtype", "float64",
            "--smoother", "newton"]
-    env = {k: v for k, v in os.environ.items()
-           if k not in ("RANK", "LOCAL_RANK", "WORLD_SIZE",
-                        "MASTER_ADDR", "MASTER_PORT", "LOCAL_WORLD_SIZE",
-                        "GROUP_RANK", "TORCHELASTIC_RUN_ID")}
-    out = subprocess.run(cmd, capture_output=True, text=True,
-                         timeout=600, cwd=repo, env=env)
-    logs = "\n".join(open(f).read() for f in glob.glob(
-        str(tmp_path / "trlogs" / "**" / "*.log"), recursive=True))
+    from tests.conftest import run_torchrun
+    out, logs = run_torchrun(cmd, repo, tmp_path / "trlogs")
     assert out.returncode == 0, (out.stderr[-1500:], logs[-1500:])
     line = [ln for ln in logs.splitlines() if ln.startswith("{")][-1]
     d = json.loads(line)
