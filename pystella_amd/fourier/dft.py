@@ -282,6 +282,7 @@ class PencilDFT(BaseDFT):
                               device=self.device)
         self.fk = torch.empty(self.shape(True), dtype=self.torch_cdtype,
                               device=self.device)
+        self._bufs = {}     # persistent all-to-all staging buffers
 
         kx = fftfreq(Nx)
         ky = fftfreq(Ny)
@@ -306,37 +307,58 @@ class PencilDFT(BaseDFT):
         return self.rank_shape
 
     # -- transpose helpers --------------------------------------------------
+    def _comm_buffers(self, n, dtype, device):
+        """Persistent flat send/recv staging buffers (per real-dtype
+        element count; transpose phases have static sizes, so two
+        buffers serve every call with zero per-call allocation)."""
+        key = (dtype, str(device))
+        bufs = self._bufs.get(key)
+        if bufs is None or bufs[0].numel() < n:
+            bufs = (torch.empty(n, dtype=dtype, device=device),
+                    torch.empty(n, dtype=dtype, device=device))
+            self._bufs[key] = bufs
+        return bufs
+
     def _all_to_all(self, chunks_in, group):
         """Exchange a list of tensors (one per peer) within ``group``;
         returns the received list (shapes from ``self._recv_shapes``).
         Handles real and complex chunks (complex moves as interleaved
-        re/im pairs — RCCL has no complex dtype)."""
+        re/im pairs — RCCL has no complex dtype).  Stages through
+        persistent buffers — no per-call allocator traffic."""
         import torch.distributed as dist
         if group is None:
             return chunks_in
         is_c = chunks_in[0].is_complex()
         w = 2 if is_c else 1
-        if is_c:
-            flat_in = [torch.view_as_real(c.contiguous()).reshape(-1)
-                       for c in chunks_in]
-        else:
-            flat_in = [c.contiguous().reshape(-1) for c in chunks_in]
-        in_sizes = [int(c.numel()) for c in flat_in]
-        send = torch.cat(flat_in)
+
+        in_sizes = [int(c.numel()) * w for c in chunks_in]
         out_sizes = [int(np.prod(s)) * w for s in self._recv_shapes]
-        recv = torch.empty(sum(out_sizes), dtype=send.dtype,
-                           device=send.device)
-        dist.all_to_all_single(recv, send, out_sizes, in_sizes, group=group)
+        rdtype = (torch.float64 if chunks_in[0].dtype
+                  in (torch.complex128, torch.float64) else torch.float32)
+        n = max(sum(in_sizes), sum(out_sizes))
+        send, recv = self._comm_buffers(n, rdtype,
+                                        chunks_in[0].device)
+        off = 0
+        for c in chunks_in:
+            src = torch.view_as_real(c) if is_c else c
+            m = int(src.numel())
+            # pack (possibly strided) split views straight into the
+            # persistent send buffer — no intermediate contiguous copy
+            send[off:off + m].view(src.shape).copy_(src)
+            off += m
+        dist.all_to_all_single(recv[:sum(out_sizes)],
+                               send[:sum(in_sizes)],
+                               out_sizes, in_sizes, group=group)
         out = []
         off = 0
         for s in self._recv_shapes:
-            n = int(np.prod(s)) * w
+            m = int(np.prod(s)) * w
             if is_c:
-                piece = recv[off:off + n].view(*s, 2)
+                piece = recv[off:off + m].view(*s, 2)
                 out.append(torch.view_as_complex(piece))
             else:
-                out.append(recv[off:off + n].view(*s))
-            off += n
+                out.append(recv[off:off + m].view(*s))
+            off += m
         return out
 
     def forward_transform(self, fx, fk):
