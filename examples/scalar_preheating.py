@@ -168,7 +168,7 @@ def main(args=None):
                     **{k: np.asarray(v) for k, v in energy.items()},
                     eos=energy["pressure"] / energy["total"],
                     constraint=expand.constraint(energy["total"]))
-                out.output("statistics_f", t=t, a=expand.a[0], **f_stats)
+                out.output("statistics/f", t=t, a=expand.a[0], **f_stats)
 
         if expand.a[0] / output.a_last_spec >= 1.05:
             output.a_last_spec = expand.a[0]

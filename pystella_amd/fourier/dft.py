@@ -25,7 +25,8 @@ from __future__ import annotations
 import numpy as np
 import torch
 
-__all__ = ["DFT", "BaseDFT", "fftfreq", "get_sliced_momenta"]
+__all__ = ["DFT", "BaseDFT", "TorchDFT", "PencilDFT", "pDFT",
+           "pyclDFT", "fftfreq", "get_sliced_momenta"]
 
 
 def fftfreq(n):
@@ -439,3 +440,9 @@ def DFT(decomp, grid_shape=None, dtype=np.float64, device="cpu", **kwargs):
     if tuple(decomp.proc_shape) == (1, 1, 1):
         return TorchDFT(decomp, grid_shape, dtype, device)
     return PencilDFT(decomp, grid_shape, dtype, device)
+
+
+# reference-API aliases (reference dft.py:352 `pDFT` = the distributed
+# transform, dft.py:430 `pyclDFT` = the single-device transform)
+pDFT = PencilDFT
+pyclDFT = TorchDFT
