@@ -277,7 +277,7 @@ class DomainDecomposition:
                                       self.rz + delta[2])
                 self._exchange_axis(fx, axis, h, (lo_rank, hi_rank))
 
-    def share_halos_start(self, fx, skip_wrap=False):
+    def share_halos_start(self, fx, skip_wrap=False, wrap_axes=None):
         """Overlap-friendly halo exchange for STAR stencils: wraps
         single-rank axes in place immediately (stream-ordered) and posts
         ALL remote-axis face exchanges in one batched non-blocking
@@ -287,6 +287,11 @@ class DomainDecomposition:
         for axis-aligned (star) stencil reads, e.g. the Laplacian hot
         loop.  Returns a handle; call ``handle.finish()`` before any
         kernel that reads the halos.
+
+        :arg wrap_axes: explicit subset of single-rank axes to wrap
+            (axes whose periodicity the consuming kernel handles
+            in-register pass a reduced set); default = all single-rank
+            axes, or none with ``skip_wrap``.
         """
         dist = _dist()
         hx, hy, hz = self.halo_shape
@@ -295,13 +300,22 @@ class DomainDecomposition:
         ops = []
         fills = []
         checked_out = []
-        wrap_axes = [] if skip_wrap else [
-            ax for ax, (h, p) in enumerate(
-                zip((hx, hy, hz), (px, py, pz))) if h > 0 and p == 1]
+        if wrap_axes is None:
+            wrap_axes = [] if skip_wrap else [
+                ax for ax, (h, p) in enumerate(
+                    zip((hx, hy, hz), (px, py, pz))) if h > 0 and p == 1]
+        else:
+            skip_wrap = True     # caller controls non-wrapped axes
         wrapped_fused = skip_wrap
         if wrap_axes and isinstance(fx, torch.Tensor) and fx.is_cuda:
             from pystella_amd.backend.hip import wrap_star
             wrap_star(fx, self.halo_shape, wrap_axes)
+            wrapped_fused = True
+        elif wrap_axes and skip_wrap:
+            # explicit subset on the CPU path
+            for ax in wrap_axes:
+                self._wrap_axis(fx, dim - 3 + ax,
+                                self.halo_shape[ax])
             wrapped_fused = True
         for ax_rel, (h, p) in enumerate(zip((hx, hy, hz), (px, py, pz))):
             if h == 0:

@@ -599,6 +599,10 @@ class DeviceFriedmannLoop:
                 self.wt[r] *= self.grid_size
                 self.wp[r] *= self.grid_size
 
+        # z-only periodic default above the all-axes volume threshold:
+        # pending A/B measurement (see _periodic_axes)
+        self._z_only_default = False
+
         # device state [a, adot, k_a, k_adot, hubble, energy, pressure]
         self.state = None
         self._expand0 = expand
@@ -675,7 +679,12 @@ class DeviceFriedmannLoop:
             env.update(self.stepper._stepper.tmp_arrays)
             kerns = self._stage_kernels(smap, env)
 
-            skip_wrap = any(kerns[0][0].periodic)
+            periodic = kerns[0][0].periodic
+            # wrap only the single-rank axes the kernels do NOT read
+            # periodically in-register
+            wrap_axes = [ax for ax, (h_, p_, per) in enumerate(zip(
+                self.decomp.halo_shape, self.decomp.proc_shape,
+                periodic)) if h_ > 0 and p_ == 1 and not per]
             # PYSTELLA_NO_OVERLAP=1: safety valve for real-xGMI bring-up
             # — sequential per-axis share_halos (no concurrent batched
             # group, corners propagated) instead of the overlapped path
@@ -685,7 +694,7 @@ class DeviceFriedmannLoop:
                 handles = []
             else:
                 handles = [self.decomp.share_halos_start(
-                               arrays[name], skip_wrap=skip_wrap)
+                               arrays[name], wrap_axes=wrap_axes)
                            for name in self.stepper.pingpong]
             interior, slabs = self._regions(kerns[0][0].rank_shape)
             if self._partials is None or \
@@ -752,15 +761,20 @@ class DeviceFriedmannLoop:
         # profiles/r02_strong_proxy.txt): in-kernel periodic reads on
         # non-decomposed axes win +14 % at 128^3 and +3.3 % at 256^3
         # per-rank (the N=8/N=64 strong-scaling shapes), are neutral at
-        # 512^3 scalar+GW.  Default: on up to 256^3-per-rank volumes;
-        # PYSTELLA_PERIODIC=1/0 forces either way.
+        # 512^3 scalar+GW.  Default: all axes up to 256^3-per-rank
+        # volumes; above that, z only (the z-wrap launch is the
+        # scattered, expensive one — r01 notes).  PYSTELLA_PERIODIC
+        # forces: 1 = all, 0 = none, or an axis subset like "z"/"yz".
         import os
+        px, py, pz = self.decomp.proc_shape
         force = os.environ.get("PYSTELLA_PERIODIC")
         if force == "0":
             return (False, False, False)
+        if force and force != "1":
+            return (px == 1 and "x" in force, py == 1 and "y" in force,
+                    pz == 1 and "z" in force)
         if force != "1" and int(np.prod(rank_shape)) > 17_000_000:
-            return (False, False, False)
-        px, py, pz = self.decomp.proc_shape
+            return (False, False, pz == 1 and self._z_only_default)
         return (px == 1, py == 1, pz == 1)
 
     def _stage_kernels(self, smap, env):
