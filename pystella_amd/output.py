@@ -1,19 +1,21 @@
 """Run output: appendable time-series datasets + provenance capture.
 
 Analogue of reference pystella/output.py:52-181 (``OutputFile`` over
-h5py).  This image ships no h5py, so the default backend is a
-self-contained directory store with the same append-mode semantics:
+h5py).  The default backend writes a REAL HDF5 file with the
+reference's layout — through h5py when importable, otherwise through
+the self-contained spec-subset writer (:mod:`pystella_amd.hdf5`; this
+image ships no h5py, and the files open with stock h5py/libhdf5
+elsewhere):
 
 * ``out.output("energy", t=..., a=..., **components)`` appends one row
-  per call to each named dataset under the group;
+  per call to each named dataset under the group (h5py-resizable
+  semantics; reference output.py:157-181);
 * run provenance (hostname, device, argv, package versions, git
-  revisions, the run script's text) is captured as attributes, matching
-  reference output.py:98-155;
-* everything is written as ``.npy`` stacks + a JSON attribute file, and
-  can be read back with :class:`OutputFile`'s ``read`` helpers.
+  revisions, the run script's text) is captured as root attributes,
+  matching reference output.py:98-155.
 
-If h5py is importable, an HDF5 file with the same layout is written
-instead (same group/dataset names and attrs).
+``PYSTELLA_OUTPUT=dir`` selects the ``.npy``-directory store instead
+(same append semantics, no binary format).
 """
 
 from __future__ import annotations
