@@ -473,8 +473,16 @@ class StencilRKStepper:
                         k_new = var(f"knew_{i}")
                         tmp[k_new] = coefA * k_acc + dtv * rhs_name
                         tmp_g[gi][k_new] = tmp[k_new]
-                        rk[k_acc] = k_new
-                        rk_g[gi][k_acc] = k_new
+                        # the LAST stage's k stores are dead: every 2N
+                        # tableau has A_0 = 0 (Williamson form), so the
+                        # next step's stage 0 multiplies the array by
+                        # zero before reading anything else.  Eliding
+                        # them removes one full k write pass per step
+                        # (PMC r02: k traffic is ~25% of family bytes).
+                        last = stage == self_inner.num_stages - 1
+                        if not last:
+                            rk[k_acc] = k_new
+                            rk_g[gi][k_acc] = k_new
                         if ff.name in _Fused.pingpong:
                             out_f = Field(f"{ff.name}_next",
                                           offset=ff.offset,
