@@ -479,7 +479,11 @@ class StencilRKStepper:
                         # zero before reading anything else.  Eliding
                         # them removes one full k write pass per step
                         # (PMC r02: k traffic is ~25% of family bytes).
-                        last = stage == self_inner.num_stages - 1
+                        # PYSTELLA_KEEP_LASTK=1 restores the stores
+                        # (A/B knob).
+                        last = (stage == self_inner.num_stages - 1
+                                and os.environ.get(
+                                    "PYSTELLA_KEEP_LASTK") != "1")
                         if not last:
                             rk[k_acc] = k_new
                             rk_g[gi][k_acc] = k_new
