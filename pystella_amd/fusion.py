@@ -482,6 +482,7 @@ class StencilRKStepper:
                         # PYSTELLA_KEEP_LASTK=1 restores the stores
                         # (A/B knob).
                         last = (stage == self_inner.num_stages - 1
+                                and float(self_inner._A[0]) == 0.0
                                 and os.environ.get(
                                     "PYSTELLA_KEEP_LASTK") != "1")
                         if not last:
