@@ -479,10 +479,16 @@ class StencilRKStepper:
                         # zero before reading anything else.  Eliding
                         # them removes one full k write pass per step
                         # (PMC r02: k traffic is ~25% of family bytes).
-                        # PYSTELLA_KEEP_LASTK=1 restores the stores
-                        # (A/B knob).
+                        # Large (nf>=4) families keep the stores: the
+                        # store-free stage-4 form recompiles worse for
+                        # the GW hij kernel (same footgun as the r01
+                        # stage-0 elision, −5% measured).
+                        # PYSTELLA_KEEP_LASTK=1 restores all stores.
+                        small_family = (ring_groups is None
+                                        or ring_groups[gi][1] < 4)
                         last = (stage == self_inner.num_stages - 1
                                 and float(self_inner._A[0]) == 0.0
+                                and small_family
                                 and os.environ.get(
                                     "PYSTELLA_KEEP_LASTK") != "1")
                         if not last:
