@@ -1618,7 +1618,8 @@ class JitLapStage:
                  scalar_names, halo, rank_shape, dx, nf, f_name="f",
                  lap_name="lap_f", name="rk_lapstage", tile=None,
                  nt=True, state_map=None, min_waves=1,
-                 periodic=(False, False, False), section_size=None):
+                 periodic=(False, False, False), section_size=None,
+                 guard_stores=frozenset(), guard_scalar=None):
         import os as _os
         from pystella_amd.field import (
             Field, Subscript, Variable, is_number, iter_exprs, walk_expr)
@@ -1902,7 +1903,27 @@ class JitLapStage:
                 lines.append(_preload_decl(nm, lin))
             _emit_tmps(tmp_items, lines)
             _entry_lines(lines)
-            _emit_stores(store_items, lines)
+
+            def _lhs_name(lhs):
+                f = lhs.aggregate if isinstance(lhs, Subscript) else lhs
+                return getattr(f, "name", None)
+
+            if guard_stores and guard_scalar is not None:
+                # runtime-skippable stores (dead last-stage k writes,
+                # see fusion.py): identical compiled form across
+                # stages, one uniform scalar branch around the block
+                normal = [(l, r) for l, r in store_items
+                          if _lhs_name(l) not in guard_stores]
+                guarded = [(l, r) for l, r in store_items
+                           if _lhs_name(l) in guard_stores]
+                _emit_stores(normal, lines)
+                if guarded:
+                    cond = cg.scalar_param(guard_scalar)
+                    lines.append(f"if ({cond} != 0.0) {{")
+                    _emit_stores(guarded, lines)
+                    lines.append("}")
+            else:
+                _emit_stores(store_items, lines)
         self.sectioned = sectioned
 
         # pointer params: stencil field first, then every other spatial
@@ -2110,12 +2131,14 @@ def get_lap_stage_kernel(map_dict, tmp_instructions, entries, field_args,
                          scalar_names, halo, rank_shape, dx, nf,
                          f_name="f", lap_name="lap_f",
                          name="rk_lapstage", tile=None, nt=True,
-                         state_map=None, periodic=(False, False, False)):
+                         state_map=None, periodic=(False, False, False),
+                         guard_stores=frozenset(), guard_scalar=None):
     return JitLapStage(map_dict, tmp_instructions, entries, field_args,
                        scalar_names, halo, rank_shape, dx, nf,
                        f_name=f_name, lap_name=lap_name, name=name,
                        tile=tile, nt=nt, state_map=state_map,
-                       periodic=periodic)
+                       periodic=periodic, guard_stores=guard_stores,
+                       guard_scalar=guard_scalar)
 
 
 def get_lap_reduction_kernel(entries, field_args, scalar_names, halo,

@@ -96,8 +96,9 @@ class _StageRedMap:
 
     def __init__(self, map_dict, tmp_instructions=None, red_entries=(),
                  reduction=None, derivs=None, lap_names=(),
-                 ring=None, **kwargs):
+                 ring=None, guard_kstores=frozenset(), **kwargs):
         from pystella_amd.elementwise import ElementWiseMap
+        self.guard_kstores = frozenset(guard_kstores)
         self._map = ElementWiseMap(map_dict, tmp_instructions, **kwargs)
         self.red_entries = list(red_entries)
         self.reduction = reduction
@@ -157,7 +158,9 @@ class _StageRedMap:
                             fargs, [], m.halo_shape, rank_shape,
                             self.derivs.dx, nf, f_name=f_name,
                             lap_name=f"lap_{f_name}",
-                            name=f"{m.name}_{f_name}")
+                            name=f"{m.name}_{f_name}",
+                            guard_stores=self.guard_kstores,
+                            guard_scalar="store_k")
                         for (rk_o, tmp_o, red_o, f_name, nf), fargs
                         in zip(self.ring, self._ring_field_args)]
                 else:
@@ -443,6 +446,7 @@ class StencilRKStepper:
                 for stage in range(self_inner.num_stages):
                     tmp = {**grad_tmps, **lap_tmps}
                     rk = {}
+                    guard_names = set()
                     ngroups = len(ring_groups) if ring_groups else 1
                     tmp_g = [dict(grad_tmps) if gi in grad_groups
                              else {} for gi in range(ngroups)]
@@ -478,22 +482,31 @@ class StencilRKStepper:
                         # next step's stage 0 multiplies the array by
                         # zero before reading anything else.  Eliding
                         # them removes one full k write pass per step
-                        # (PMC r02: k traffic is ~25% of family bytes).
-                        # Large (nf>=4) families keep the stores: the
-                        # store-free stage-4 form recompiles worse for
-                        # the GW hij kernel (same footgun as the r01
-                        # stage-0 elision, −5% measured).
+                        # (PMC r02: k traffic is ~25% of family bytes;
+                        # measured +4.4% flagship, r02_laststage A/B).
+                        # Small families drop the store structurally;
+                        # large (nf>=4) families keep a UNIFORM
+                        # compiled form across stages (the store-free
+                        # stage-4 form recompiles worse for the GW hij
+                        # kernel, −5% — same footgun as the r01
+                        # stage-0 elision) and instead guard the k
+                        # stores behind the runtime `store_k` scalar,
+                        # 0.0 at the last stage.
                         # PYSTELLA_KEEP_LASTK=1 restores all stores.
+                        keep_env = os.environ.get(
+                            "PYSTELLA_KEEP_LASTK") == "1"
                         small_family = (ring_groups is None
                                         or ring_groups[gi][1] < 4)
                         last = (stage == self_inner.num_stages - 1
                                 and float(self_inner._A[0]) == 0.0
-                                and small_family
-                                and os.environ.get(
-                                    "PYSTELLA_KEEP_LASTK") != "1")
-                        if not last:
+                                and not keep_env)
+                        if small_family and last:
+                            pass            # structural elision
+                        else:
                             rk[k_acc] = k_new
                             rk_g[gi][k_acc] = k_new
+                            if not small_family and not keep_env:
+                                guard_names.add(f"{ff.name}_tmp")
                         if ff.name in _Fused.pingpong:
                             out_f = Field(f"{ff.name}_next",
                                           offset=ff.offset,
@@ -517,11 +530,16 @@ class StencilRKStepper:
                                 in enumerate(ring_groups)]
                         fp = dict(fixed_parameters or {})
                         fp.setdefault("rk_a0", 0.0)
+                        is_last = (stage == self_inner.num_stages - 1
+                                   and float(self_inner._A[0]) == 0.0)
+                        fp["store_k"] = 0.0 if (guard_names and is_last) \
+                            else 1.0
                         steps.append(_StageRedMap(
                             rk, tmp_instructions=tmp,
                             red_entries=red_entries,
                             reduction=reduction, derivs=derivs_ref,
                             lap_names=lap_name_list, ring=ring,
+                            guard_kstores=frozenset(guard_names),
                             halo_shape=self_inner.halo_shape,
                             rank_shape=self_inner.rank_shape,
                             name=f"rk_stage_red{stage}",
@@ -696,6 +714,8 @@ class DeviceFriedmannLoop:
                 self.stepper._stepper.tmp_arrays = \
                     self.stepper._stepper.get_tmp_arrays_like(**arrays)
             env.update(self.stepper._stepper.tmp_arrays)
+            env["store_k"] = smap._map.fixed_parameters.get(
+                "store_k", 1.0)
             kerns = self._stage_kernels(smap, env)
 
             periodic = kerns[0][0].periodic
@@ -818,7 +838,9 @@ class DeviceFriedmannLoop:
                     f_name=f_name, lap_name=f"lap_{f_name}",
                     name=f"{m.name}_{f_name}",
                     state_map={"a": 0, "hubble": 4},
-                    periodic=periodic)
+                    periodic=periodic,
+                    guard_stores=smap.guard_kstores,
+                    guard_scalar="store_k")
                 for (rk_o, tmp_o, red_o, f_name, nf), fargs
                 in zip(smap.ring, smap._ring_field_args)]
             smap._hip_kernel = kerns
