@@ -482,16 +482,15 @@ class StencilRKStepper:
                         # next step's stage 0 multiplies the array by
                         # zero before reading anything else.  Eliding
                         # them removes one full k write pass per step
-                        # (PMC r02: k traffic is ~25% of family bytes;
-                        # measured +4.4% flagship, r02_laststage A/B).
-                        # Small families drop the store structurally;
-                        # large (nf>=4) families keep a UNIFORM
-                        # compiled form across stages (the store-free
-                        # stage-4 form recompiles worse for the GW hij
-                        # kernel, −5% — same footgun as the r01
-                        # stage-0 elision) and instead guard the k
-                        # stores behind the runtime `store_k` scalar,
-                        # 0.0 at the last stage.
+                        # for SMALL families: measured +4.4 % flagship
+                        # (profiles/r02_laststage_elision_ab.txt).
+                        # Large (nf>=4) families keep unconditional
+                        # stores: BOTH alternatives measured worse for
+                        # the GW hij kernel — the store-free stage-4
+                        # form recompiles slower (−5 %), and runtime-
+                        # guarded stores (`store_k` branch, uniform
+                        # form) collapse it to a 150-VGPR serialized
+                        # form at −40 % (profiles/r02_guard_ab.txt).
                         # PYSTELLA_KEEP_LASTK=1 restores all stores.
                         keep_env = os.environ.get(
                             "PYSTELLA_KEEP_LASTK") == "1"
@@ -500,13 +499,9 @@ class StencilRKStepper:
                         last = (stage == self_inner.num_stages - 1
                                 and float(self_inner._A[0]) == 0.0
                                 and not keep_env)
-                        if small_family and last:
-                            pass            # structural elision
-                        else:
+                        if not (small_family and last):
                             rk[k_acc] = k_new
                             rk_g[gi][k_acc] = k_new
-                            if not small_family and not keep_env:
-                                guard_names.add(f"{ff.name}_tmp")
                         if ff.name in _Fused.pingpong:
                             out_f = Field(f"{ff.name}_next",
                                           offset=ff.offset,
