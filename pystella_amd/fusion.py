@@ -733,12 +733,17 @@ class DeviceFriedmannLoop:
                 handles = []
                 split_axes = None
             else:
-                # run the periodic-wrap kernels on a side stream,
-                # concurrent with the interior launch (the boundary
-                # slabs wait on the wrap event); measured ~0.6 ms/step
-                # of serial wrap time at 512^3 single-GPU
+                # OPT-IN (PYSTELLA_WRAP_OVERLAP=1): periodic-wrap
+                # kernels on a side stream under the interior launch.
+                # Measured −29 % at 512³ (profiles/
+                # r02_wrapoverlap_ab.txt): the thin z boundary slabs
+                # run at ~1/32 lane efficiency under the (64z,8y)
+                # tile, costing far more than the 0.6 ms/step of
+                # serial wrap time the overlap reclaims.  Kept as a
+                # knob; a per-slab tile shape would be needed to make
+                # it pay.
                 if wrap_axes and f.is_cuda and os.environ.get(
-                        "PYSTELLA_WRAP_OVERLAP") != "0":
+                        "PYSTELLA_WRAP_OVERLAP") == "1":
                     import torch as _t
                     from pystella_amd.backend.hip import wrap_star
                     if self._side_stream is None:
