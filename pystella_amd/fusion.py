@@ -773,11 +773,17 @@ class DeviceFriedmannLoop:
                         for ax, p_ in enumerate((px_, py_, pz_)))
             interior, slabs = self._regions(kerns[0][0].rank_shape,
                                             split_axes)
+            # thin boundary slabs get tile-matched kernel variants
+            # (a z-slab of thickness h under the (64z, 8y) main tile
+            # would run at h/64 lane efficiency)
+            slab_kerns = [self._slab_kerns(smap, kerns, b)
+                          for b in slabs]
             if self._partials is None or \
                     self._boxes != (interior, tuple(slabs)):
                 self._boxes = (interior, tuple(slabs))
-                nblks = [kerns[0][0].box_nblk(b)
-                         for b in (interior, *slabs)]
+                nblks = [kerns[0][0].box_nblk(interior)] + [
+                    bk[0][0].box_nblk(b)
+                    for bk, b in zip(slab_kerns, slabs)]
                 self._nblks = nblks
                 self._nblk_tot = sum(nblks)
                 # one partials buffer per kernel family; only the
@@ -802,8 +808,9 @@ class DeviceFriedmannLoop:
                 import torch as _t
                 _t.cuda.current_stream().wait_event(wrap_event)
             bid0 = self._nblks[0]
-            for slab, nb in zip(slabs, self._nblks[1:]):
-                for (kern, _), partials in zip(kerns, self._partials):
+            for slab, bkerns, nb in zip(slabs, slab_kerns,
+                                        self._nblks[1:]):
+                for (kern, _), partials in zip(bkerns, self._partials):
                     kern.launch_box(env, slab, partials, bid0,
                                     self._nblk_tot)
                 bid0 += nb
@@ -856,6 +863,41 @@ class DeviceFriedmannLoop:
             return (False, False, pz == 1 and self._z_only_default)
         return (px == 1, py == 1, pz == 1)
 
+    def _slab_kerns(self, smap, kerns, box):
+        """Tile-matched kernel variants for thin boundary slabs: the
+        main (64z, 8y) tile leaves a thickness-h z-slab at h/64 lane
+        efficiency (and a y-slab at h/8); variants re-tile so every
+        lane maps to a real site.  PYSTELLA_SLAB_TILES=0 disables."""
+        if os.environ.get("PYSTELLA_SLAB_TILES") == "0":
+            return kerns
+        builder = getattr(smap, "_ring_builder", None)
+        if builder is None:
+            return kerns
+        i0, i1, j0, j1, k0, k1 = box
+        tbz, tby, xch = kerns[0][0].tile
+        kz, jy = k1 - k0, j1 - j0
+        if kz < tbz and kz <= 8 and kz <= jy:
+            # next power of two >= kz keeps the block a multiple of 64
+            tz = 1 << (max(1, kz) - 1).bit_length()
+            tile = (tz, max(1, 256 // tz), xch)
+            key = ("z", tile)
+        elif jy < tby and jy <= 8:
+            ty = 1 << (max(1, jy) - 1).bit_length()
+            tile = (tbz, ty, xch)
+            key = ("y", tile)
+        else:
+            return kerns
+        variants = getattr(smap, "_slab_variants", None)
+        if variants is None:
+            smap._slab_variants = variants = {}
+        kv = variants.get(key)
+        if kv is None:
+            built = builder(tile=tile, suffix=f"_{key[0]}slab")
+            has_red = [bool(r[2]) for r in smap.ring]
+            kv = list(zip(built, has_red))
+            variants[key] = kv
+        return kv
+
     def _stage_kernels(self, smap, env):
         """List of (ring kernel, has_reducers) for this stage, compiled
         with the device-state scalar map."""
@@ -871,20 +913,26 @@ class DeviceFriedmannLoop:
         if not ok:
             from pystella_amd.backend.hip import get_lap_stage_kernel
             periodic = self._periodic_axes(smap, rank_shape)
-            kerns = [
-                get_lap_stage_kernel(
-                    rk_o, tmp_o, red_o or [(0.0, "sum")], fargs, [],
-                    m.halo_shape, rank_shape, smap.derivs.dx, nf,
-                    f_name=f_name, lap_name=f"lap_{f_name}",
-                    name=f"{m.name}_{f_name}",
-                    state_map={"a": 0, "hubble": 4},
-                    periodic=periodic,
-                    guard_stores=smap.guard_kstores,
-                    guard_scalar="store_k")
-                for (rk_o, tmp_o, red_o, f_name, nf), fargs
-                in zip(smap.ring, smap._ring_field_args)]
+
+            def build(tile=None, suffix=""):
+                return [
+                    get_lap_stage_kernel(
+                        rk_o, tmp_o, red_o or [(0.0, "sum")], fargs, [],
+                        m.halo_shape, rank_shape, smap.derivs.dx, nf,
+                        f_name=f_name, lap_name=f"lap_{f_name}",
+                        name=f"{m.name}_{f_name}{suffix}",
+                        state_map={"a": 0, "hubble": 4},
+                        periodic=periodic, tile=tile,
+                        guard_stores=smap.guard_kstores,
+                        guard_scalar="store_k")
+                    for (rk_o, tmp_o, red_o, f_name, nf), fargs
+                    in zip(smap.ring, smap._ring_field_args)]
+
+            kerns = build()
             smap._hip_kernel = kerns
             smap._kernel_shape = rank_shape
+            smap._ring_builder = build
+            smap._slab_variants = {}
         has_red = [bool(r[2]) for r in smap.ring]
         return list(zip(kerns, has_red))
 

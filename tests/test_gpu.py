@@ -1031,7 +1031,7 @@ def test_device_loop_region_split_gpu(grid_shape=(32, 32, 32)):
         if split:
             nx, ny, nz = grid_shape
 
-            def fake_regions(rank_shape):
+            def fake_regions(rank_shape, split_axes=None):
                 return ((0, nx, 0, ny, h, nz - h),
                         [(0, nx, 0, ny, 0, h),
                          (0, nx, 0, ny, nz - h, nz)])
