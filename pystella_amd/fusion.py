@@ -635,6 +635,7 @@ class DeviceFriedmannLoop:
         # pending A/B measurement (see _periodic_axes)
         self._z_only_default = False
         self._side_stream = None    # periodic-wrap overlap stream
+        self._slab_streams = None   # concurrent boundary-slab streams
 
         # device state [a, adot, k_a, k_adot, hubble, energy, pressure]
         self.state = None
@@ -807,13 +808,47 @@ class DeviceFriedmannLoop:
             if wrap_event is not None:
                 import torch as _t
                 _t.cuda.current_stream().wait_event(wrap_event)
+            # Boundary slabs are latency-bound (each launch has ~1
+            # wave/CU); run them on CONCURRENT side streams so their
+            # waves co-reside (and overlap the interior tail) instead
+            # of executing sequentially — measured 2.3 ms/step of
+            # serial slab time at the 256^3 N=8-rank proxy.
+            # PYSTELLA_SLAB_STREAMS=0 falls back to in-order launches.
+            use_streams = (slabs and f.is_cuda and os.environ.get(
+                "PYSTELLA_SLAB_STREAMS") != "0")
+            if use_streams:
+                import torch as _t
+                if self._slab_streams is None:
+                    self._slab_streams = [_t.cuda.Stream()
+                                          for _ in range(6)]
+                ev_ready = _t.cuda.Event()
+                ev_ready.record()
+                slab_evs = []
             bid0 = self._nblks[0]
-            for slab, bkerns, nb in zip(slabs, slab_kerns,
-                                        self._nblks[1:]):
-                for (kern, _), partials in zip(bkerns, self._partials):
-                    kern.launch_box(env, slab, partials, bid0,
-                                    self._nblk_tot)
+            for idx, (slab, bkerns, nb) in enumerate(
+                    zip(slabs, slab_kerns, self._nblks[1:])):
+                if use_streams:
+                    stream = self._slab_streams[
+                        idx % len(self._slab_streams)]
+                    with _t.cuda.stream(stream):
+                        stream.wait_event(ev_ready)
+                        for (kern, _), partials in zip(
+                                bkerns, self._partials):
+                            kern.launch_box(env, slab, partials, bid0,
+                                            self._nblk_tot)
+                        e = _t.cuda.Event()
+                        e.record()
+                        slab_evs.append(e)
+                else:
+                    for (kern, _), partials in zip(bkerns,
+                                                   self._partials):
+                        kern.launch_box(env, slab, partials, bid0,
+                                        self._nblk_tot)
                 bid0 += nb
+            if use_streams:
+                cur = _t.cuda.current_stream()
+                for e in slab_evs:
+                    cur.wait_event(e)
 
             if self._fk is None:
                 from pystella_amd.backend.hip import JitFriedmann
@@ -864,11 +899,13 @@ class DeviceFriedmannLoop:
         return (px == 1, py == 1, pz == 1)
 
     def _slab_kerns(self, smap, kerns, box):
-        """Tile-matched kernel variants for thin boundary slabs: the
-        main (64z, 8y) tile leaves a thickness-h z-slab at h/64 lane
-        efficiency (and a y-slab at h/8); variants re-tile so every
-        lane maps to a real site.  PYSTELLA_SLAB_TILES=0 disables."""
-        if os.environ.get("PYSTELLA_SLAB_TILES") == "0":
+        """Tile-matched kernel variants for thin boundary slabs
+        (OPT-IN, PYSTELLA_SLAB_TILES=1): re-tiles so every lane maps
+        to a real site.  Measured SLOWER than the main tile at the
+        256^3 proxy (3447 vs 3982; scattered access beats idle lanes
+        here) — the slab cost is latency, addressed by the concurrent
+        slab streams below instead."""
+        if os.environ.get("PYSTELLA_SLAB_TILES") != "1":
             return kerns
         builder = getattr(smap, "_ring_builder", None)
         if builder is None:
