@@ -1533,6 +1533,97 @@ class _NTLapCodegen(_LapCodegen):
         return super().field_access(f, outer_idx)
 
 
+# Multi-box "shell" form: ONE launch covers all boundary slabs of a
+# rank (separate thin-slab launches are latency-bound at ~1 wave/CU
+# each and serialize).  Box bounds and per-box block-grid tables are
+# baked as constant arrays; blocks decode their (box, tile) from the
+# flat blockIdx.x, and the reduction tail indexes partials by the flat
+# block id.
+LAPSTAGE_SHELL_TEMPLATE = """{defines}
+{preamble}
+#define NRED {nred}
+#define NF {nf}
+#define NBOX {nbox}
+__constant__ int sh_k0[NBOX] = {{ {k0s} }};
+__constant__ int sh_k1[NBOX] = {{ {k1s} }};
+__constant__ int sh_j0[NBOX] = {{ {j0s} }};
+__constant__ int sh_j1[NBOX] = {{ {j1s} }};
+__constant__ int sh_i0[NBOX] = {{ {i0s} }};
+__constant__ int sh_i1[NBOX] = {{ {i1s} }};
+__constant__ int sh_gz[NBOX] = {{ {gzs} }};
+__constant__ int sh_gy[NBOX] = {{ {gys} }};
+__constant__ int sh_blk0[NBOX] = {{ {blk0s} }};
+extern "C" __global__ __launch_bounds__(TBZ * TBY, MINW) void {name}(
+    {params})
+{{
+    double acc[NRED];
+    {init}
+    const int bflat = (int)blockIdx.x;
+    int box = 0;
+    #pragma unroll
+    for (int b = 1; b < NBOX; ++b) box += (bflat >= sh_blk0[b]);
+    const int lb = bflat - sh_blk0[box];
+    const int bz = lb % sh_gz[box];
+    const int by = (lb / sh_gz[box]) % sh_gy[box];
+    const int bx = lb / (sh_gz[box] * sh_gy[box]);
+    const int k = sh_k0[box] + bz * TBZ + (int)(threadIdx.x % TBZ);
+    const int j = sh_j0[box] + by * TBY + (int)(threadIdx.x / TBZ);
+    const int i0 = sh_i0[box] + bx * XCHUNK;
+    const int i1 = (i0 + XCHUNK < sh_i1[box]) ? i0 + XCHUNK
+                                              : sh_i1[box];
+    if (k < sh_k1[box] && j < sh_j1[box]) {{
+        const long sx = PSY * PSZ;
+        {wrap_decls}
+        double lapv[NF];
+        double ring[NF][2 * H + 1];
+        #pragma unroll
+        for (int fld = 0; fld < NF; ++fld) {{
+            const double* fp = {fname} + (long)fld * PVOL
+                               + (long)(j + H) * PSZ + (k + H);
+            #pragma unroll
+            for (int p = 0; p < 2 * H; ++p) {{
+                {ring_init}
+            }}
+        }}
+        for (int i = i0; i < i1; ++i) {{
+            {x_off}
+            #pragma unroll
+            for (int fld = 0; fld < NF; ++fld) {{
+                const double* fp = {fname} + (long)fld * PVOL
+                                   + (long)(j + H) * PSZ + (k + H);
+                {ring_load}
+                const double* cp = fp + (long)(i + H) * sx;
+                double la = ring[fld][H] * LAPC0;
+                {lap_terms}
+                lapv[fld] = la;
+            }}
+            {body}
+            #pragma unroll
+            for (int fld = 0; fld < NF; ++fld)
+                #pragma unroll
+                for (int p = 0; p < 2 * H; ++p)
+                    ring[fld][p] = ring[fld][p + 1];
+        }}
+    }}
+    __shared__ double sd[TBZ * TBY];
+    const int bid = bid0 + bflat;
+    for (int r = 0; r < NRED; ++r) {{
+        sd[threadIdx.x] = acc[r];
+        __syncthreads();
+        for (int s = (TBZ * TBY) / 2; s > 0; s >>= 1) {{
+            if ((int)threadIdx.x < s)
+                sd[threadIdx.x] = COMBINE(r, sd[threadIdx.x],
+                                          sd[threadIdx.x + s]);
+            __syncthreads();
+        }}
+        if (threadIdx.x == 0)
+            partials[(long)r * nblkT + bid] = sd[0];
+        __syncthreads();
+    }}
+}}
+"""
+
+
 LAPSTAGE_SEC_TEMPLATE = """{defines}
 {preamble}
 #define NRED {nred}
@@ -1961,8 +2052,9 @@ class JitLapStage:
                 wrap_decls="\n        ".join(wrap_decls),
                 x_off=x_off, ring_init=ring_init,
                 site_body="\n            ".join(lines))
+            self._shell_parts = None
         else:
-            src = LAPSTAGE_TEMPLATE.format(
+            fmt = dict(
                 defines=defines, preamble=PREAMBLE, nred=len(entries),
                 nf=nf, name=name, params=params, fname=f_name,
                 init="\n    ".join(init_lines),
@@ -1970,6 +2062,9 @@ class JitLapStage:
                 x_off=x_off, ring_load=ring_load, ring_init=ring_init,
                 lap_terms="\n                ".join(lap_terms),
                 body="\n            ".join(lines))
+            src = LAPSTAGE_TEMPLATE.format(**fmt)
+            self._shell_parts = fmt
+        self._shell_cache = {}
         self.source = src
         self.scalar_keys = [k for _, k in cg.scalars]
         self.key = ext().jit_compile(src, name)
@@ -1993,6 +2088,53 @@ class JitLapStage:
     def box_nblk(self, box):
         grid, _ = self._box_geometry(box)
         return grid[0] * grid[1] * grid[2]
+
+    # -- multi-box shell launch (all boundary slabs in ONE kernel) ------
+    def _get_shell(self, boxes):
+        cached = self._shell_cache.get(boxes)
+        if cached is not None:
+            return cached
+        if self._shell_parts is None:
+            raise RuntimeError("shell form unavailable (sectioned)")
+        grids = [self._box_geometry(b)[0] for b in boxes]
+        nbs = [g[0] * g[1] * g[2] for g in grids]
+        blk0 = [0]
+        for n in nbs[:-1]:
+            blk0.append(blk0[-1] + n)
+        fmt = dict(self._shell_parts)
+        fmt["name"] = fmt["name"] + "_shell"
+        fmt["nbox"] = len(boxes)
+
+        def ints(vals):
+            return ", ".join(str(int(v)) for v in vals)
+
+        src = LAPSTAGE_SHELL_TEMPLATE.format(
+            k0s=ints(b[4] for b in boxes), k1s=ints(b[5] for b in boxes),
+            j0s=ints(b[2] for b in boxes), j1s=ints(b[3] for b in boxes),
+            i0s=ints(b[0] for b in boxes), i1s=ints(b[1] for b in boxes),
+            gzs=ints(g[0] for g in grids), gys=ints(g[1] for g in grids),
+            blk0s=ints(blk0), **fmt)
+        kid = ext().jit_compile(src, fmt["name"])
+        cached = (kid, sum(nbs))
+        self._shell_cache[boxes] = cached
+        return cached
+
+    def shell_nblk(self, boxes):
+        return self._get_shell(tuple(boxes))[1]
+
+    def launch_shell(self, env, boxes, partials, bid0, nblk_tot):
+        """ONE launch covering every box in ``boxes`` (the rank's
+        boundary slabs); per-block partials land flat at ``bid0``."""
+        kid, total = self._get_shell(tuple(boxes))
+        ptrs = []
+        for n in self.ptr_names:
+            t = _check_tensor(n, env[n])
+            ptrs.append(t.data_ptr())
+        doubles = [_resolve_scalar(env, k) for k in self.scalar_keys]
+        ext().jit_launch(kid, total, 1, 1, self.block, 1, 1,
+                         self.shmem, _stream(),
+                         ptrs + [partials.data_ptr()],
+                         [0, 0, 0, 0, 0, 0, nblk_tot, bid0], doubles)
 
     def launch_box(self, env, box, partials, bid0, nblk_tot):
         """Launch over a sub-box, writing this launch's per-block

@@ -774,17 +774,28 @@ class DeviceFriedmannLoop:
                         for ax, p_ in enumerate((px_, py_, pz_)))
             interior, slabs = self._regions(kerns[0][0].rank_shape,
                                             split_axes)
-            # thin boundary slabs get tile-matched kernel variants
-            # (a z-slab of thickness h under the (64z, 8y) main tile
-            # would run at h/64 lane efficiency)
-            slab_kerns = [self._slab_kerns(smap, kerns, b)
-                          for b in slabs]
+            # default: ONE "shell" launch covers all boundary slabs
+            # (separate thin-slab launches are latency-bound at ~1
+            # wave/CU each and serialize — measured 2.3 ms/step of
+            # slab time at the 256^3 N=8-rank proxy).
+            # PYSTELLA_SHELL=0 falls back to per-slab launches.
+            use_shell = (bool(slabs) and f.is_cuda
+                         and os.environ.get("PYSTELLA_SHELL") != "0"
+                         and getattr(kerns[0][0], "_shell_parts", None)
+                         is not None)
+            slab_kerns = ([] if use_shell
+                          else [self._slab_kerns(smap, kerns, b)
+                                for b in slabs])
             if self._partials is None or \
                     self._boxes != (interior, tuple(slabs)):
                 self._boxes = (interior, tuple(slabs))
-                nblks = [kerns[0][0].box_nblk(interior)] + [
-                    bk[0][0].box_nblk(b)
-                    for bk, b in zip(slab_kerns, slabs)]
+                if use_shell:
+                    nblks = [kerns[0][0].box_nblk(interior),
+                             kerns[0][0].shell_nblk(slabs)]
+                else:
+                    nblks = [kerns[0][0].box_nblk(interior)] + [
+                        bk[0][0].box_nblk(b)
+                        for bk, b in zip(slab_kerns, slabs)]
                 self._nblks = nblks
                 self._nblk_tot = sum(nblks)
                 # one partials buffer per kernel family; only the
@@ -814,6 +825,11 @@ class DeviceFriedmannLoop:
             # of executing sequentially — measured 2.3 ms/step of
             # serial slab time at the 256^3 N=8-rank proxy.
             # PYSTELLA_SLAB_STREAMS=0 falls back to in-order launches.
+            if use_shell:
+                for (kern, _), partials in zip(kerns, self._partials):
+                    kern.launch_shell(env, slabs, partials,
+                                      self._nblks[0], self._nblk_tot)
+                slabs = []      # handled; skip the per-slab paths
             use_streams = (slabs and f.is_cuda and os.environ.get(
                 "PYSTELLA_SLAB_STREAMS") != "0")
             if use_streams:
