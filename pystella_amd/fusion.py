@@ -636,6 +636,7 @@ class DeviceFriedmannLoop:
         self._z_only_default = False
         self._side_stream = None    # periodic-wrap overlap stream
         self._slab_streams = None   # concurrent boundary-slab streams
+        self._shell_stream = None   # shell-kernel overlap stream
 
         # device state [a, adot, k_a, k_adot, hubble, energy, pressure]
         self.state = None
@@ -825,10 +826,32 @@ class DeviceFriedmannLoop:
             # of executing sequentially — measured 2.3 ms/step of
             # serial slab time at the 256^3 N=8-rank proxy.
             # PYSTELLA_SLAB_STREAMS=0 falls back to in-order launches.
+            shell_ev = None
             if use_shell:
-                for (kern, _), partials in zip(kerns, self._partials):
-                    kern.launch_shell(env, slabs, partials,
-                                      self._nblks[0], self._nblk_tot)
+                import torch as _t
+                if os.environ.get("PYSTELLA_SHELL") == "sync":
+                    for (kern, _), partials in zip(kerns,
+                                                   self._partials):
+                        kern.launch_shell(env, slabs, partials,
+                                          self._nblks[0],
+                                          self._nblk_tot)
+                else:
+                    # the shell depends only on the fresh halos, not
+                    # the interior launch — run it on a side stream so
+                    # it overlaps the interior's tail
+                    if self._shell_stream is None:
+                        self._shell_stream = _t.cuda.Stream()
+                    ev_ready = _t.cuda.Event()
+                    ev_ready.record()
+                    with _t.cuda.stream(self._shell_stream):
+                        self._shell_stream.wait_event(ev_ready)
+                        for (kern, _), partials in zip(
+                                kerns, self._partials):
+                            kern.launch_shell(env, slabs, partials,
+                                              self._nblks[0],
+                                              self._nblk_tot)
+                        shell_ev = _t.cuda.Event()
+                        shell_ev.record()
                 slabs = []      # handled; skip the per-slab paths
             use_streams = (slabs and f.is_cuda and os.environ.get(
                 "PYSTELLA_SLAB_STREAMS") != "0")
@@ -865,6 +888,9 @@ class DeviceFriedmannLoop:
                 cur = _t.cuda.current_stream()
                 for e in slab_evs:
                     cur.wait_event(e)
+            if shell_ev is not None:
+                import torch as _t
+                _t.cuda.current_stream().wait_event(shell_ev)
 
             if self._fk is None:
                 from pystella_amd.backend.hip import JitFriedmann
