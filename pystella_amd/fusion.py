@@ -829,7 +829,11 @@ class DeviceFriedmannLoop:
             shell_ev = None
             if use_shell:
                 import torch as _t
-                if os.environ.get("PYSTELLA_SHELL") == "sync":
+                if os.environ.get("PYSTELLA_SHELL") != "stream":
+                    # in-order launch (default): at these shapes the
+                    # interior saturates the GPU, so the side-stream
+                    # form measured no better and costs two events
+                    # (profiles/r02_shell_evolution.txt)
                     for (kern, _), partials in zip(kerns,
                                                    self._partials):
                         kern.launch_shell(env, slabs, partials,
