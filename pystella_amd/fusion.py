@@ -730,10 +730,14 @@ class DeviceFriedmannLoop:
             # group, corners propagated) instead of the overlapped path
             wrap_event = None
             if os.environ.get("PYSTELLA_NO_OVERLAP") == "1":
+                # halos are fully fresh here, so the region split is
+                # pointless — ONE full-box launch (this also makes
+                # NO_OVERLAP the true no-split alternative strategy:
+                # serial comm + peak-efficiency compute)
                 for name in self.stepper.pingpong:
                     self.decomp.share_halos(arrays[name])
                 handles = []
-                split_axes = None
+                split_axes = (False, False, False)
             else:
                 # OPT-IN (PYSTELLA_WRAP_OVERLAP=1): periodic-wrap
                 # kernels on a side stream under the interior launch.
