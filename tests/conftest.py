@@ -49,7 +49,7 @@ def _dist_worker(rank, world_size, init_file, fn, args):
         dist.destroy_process_group()
 
 
-def run_torchrun(cmd, cwd, log_dir, attempts=2, timeout=600):
+def run_torchrun(cmd, cwd, log_dir, attempts=3, timeout=600):
     """Run a torchrun command, retrying once on failure (multi-process
     rendezvous is occasionally flaky under CI load); returns
     (CompletedProcess, combined log text)."""
