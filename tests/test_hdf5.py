@@ -183,3 +183,17 @@ def test_h5py_parity_if_available(tmp_path):
     with h5py.File(path, "r") as hf:
         assert hf.attrs["argv"] in (b"x", "x")
         assert np.array_equal(hf["energy"]["total"][:], np.arange(3.0))
+
+
+def test_dir_store_backend(tmp_path, monkeypatch):
+    """PYSTELLA_OUTPUT=dir selects the .npy-directory store (same
+    append semantics, no binary format)."""
+    import pystella_amd as ps
+    monkeypatch.setenv("PYSTELLA_OUTPUT", "dir")
+    os.chdir(tmp_path)
+    out = ps.OutputFile(name="druns")
+    out.output("energy", t=0.0, total=2.5)
+    out.output("energy", t=0.1, total=2.4)
+    assert np.allclose(out.read("energy", "total"), [2.5, 2.4])
+    assert os.path.isdir("druns")
+    assert os.path.exists("druns/attrs.json")
