@@ -212,3 +212,29 @@ def _preheating_gws_8rank_worker(rank, world_size):
 def test_scalar_preheating_gws_8rank():
     from tests.conftest import run_distributed
     run_distributed(_preheating_gws_8rank_worker, 8)
+
+
+def test_bench_4rank_gws_cpu(tmp_path):
+    """4-rank (2,2,1) gloo bench with the GW tensor sector: multi-
+    family ring path + overlapped exchange at a 2-D topology."""
+    import glob
+    import json
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [sys.executable, "-m", "torch.distributed.run",
+           "--master-addr", "127.0.0.1", "--master-port", str(port),
+           "--nnodes=1", "--nproc-per-node", "4",
+           "--redirects", "3", "--log-dir", str(tmp_path / "trlogs"),
+           os.path.join(repo, "bench.py"),
+           "--gpus", "4", "--steps", "2", "--warmup", "1",
+           "--grid", "16", "--device", "cpu", "--gws"]
+    from tests.conftest import run_torchrun
+    out, logs = run_torchrun(cmd, repo, tmp_path / "trlogs")
+    assert out.returncode == 0, (out.stderr[-1500:], logs[-1500:])
+    line = [ln for ln in logs.splitlines() if ln.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["config"]["parallelism"] == "decomp3d[2, 2, 1]"
+    assert d["config"]["model"] == "scalar_preheating+gw"
