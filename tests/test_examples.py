@@ -238,3 +238,14 @@ def test_bench_4rank_gws_cpu(tmp_path):
     d = json.loads(line)
     assert d["config"]["parallelism"] == "decomp3d[2, 2, 1]"
     assert d["config"]["model"] == "scalar_preheating+gw"
+
+
+def test_tutorial(tmp_path):
+    """The runnable tutorial (examples/tutorial.py, the executable
+    form of docs/TUTORIAL.md) covers every layer end to end."""
+    import tutorial
+    os.chdir(tmp_path)
+    energy = tutorial.main(["--device", "cpu", "--n", "12"])
+    assert np.isfinite(energy["total"])
+    assert os.path.exists("tutorial_out.h5")
+    assert os.path.exists("tutorial_ckpt.h5")
