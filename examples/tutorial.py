@@ -170,7 +170,11 @@ def main(args=None):
                     {"f": arrays["f"]}, attrs={"t": dt})
     f_back = torch.zeros_like(arrays["f"])
     attrs = load_checkpoint("tutorial_ckpt.h5", decomp, {"f": f_back})
-    assert torch.equal(f_back, arrays["f"])
+    # compare interiors: the device loop leaves halos stale by design
+    # (they are refreshed lazily at the next stage's exchange), while
+    # the checkpoint restore re-shares them
+    cut = (slice(None),) + (slice(h, -h),) * 3
+    assert torch.equal(f_back[cut], arrays["f"][cut])
     print("8. wrote", out.filename, "and tutorial_ckpt.h5; restored",
           "t =", attrs["t"])
     print("tutorial complete")
