@@ -23,7 +23,8 @@ import torch
 
 from pystella_amd.reduction import Reduction
 
-__all__ = ["FusedLaplacianReduction"]
+__all__ = ["FusedLaplacianReduction", "StencilRKStepper",
+           "DeviceFriedmannLoop"]
 
 
 class FusedLaplacianReduction(Reduction):
