@@ -1248,3 +1248,67 @@ def test_device_loop_noncubic_split_gpu(grid_shape=(16, 32, 24)):
     assert torch.equal(f_one[cut], f_spl[cut])
     assert st_one["a"] == st_spl["a"]
     assert st_one["energy"] == st_spl["energy"]
+
+
+@requires_gpu
+def test_device_loop_full_shell_split_gpu(grid_shape=(24, 24, 24)):
+    """Force the FULL 6-slab split (the (2,2,2) N=8 rank geometry):
+    the multi-box shell kernel's flat-block decode across mixed x/y/z
+    slab orientations must reproduce the single-launch result
+    bit-exactly."""
+    from pystella_amd.fusion import (
+        DeviceFriedmannLoop, FusedLaplacianReduction, StencilRKStepper)
+    from pystella_amd.sectors import get_rho_and_p
+    h = 2
+    decomp = ps.DomainDecomposition((1, 1, 1), h, rank_shape=grid_shape)
+    dx = (0.3, 0.31, 0.29)
+    dt = 0.005
+    gsize = float(np.prod(grid_shape))
+    sector = ps.ScalarSector(2, potential=lambda f: f[0]**2 / 2)
+    derivs = ps.FiniteDifferencer(decomp, h, dx, rank_shape=grid_shape)
+    pad = tuple(n + 2 * h for n in grid_shape)
+    torch.manual_seed(81)
+    f0 = 0.2 + 0.01 * torch.rand((2,) + pad, dtype=torch.float64)
+    d0 = 0.01 * torch.rand((2,) + pad, dtype=torch.float64)
+
+    def run(split):
+        st = StencilRKStepper(ps.LowStorageRK54, [sector], derivs,
+                              halo_shape=h, rank_shape=grid_shape,
+                              dt=dt, reducers=sector, grid_size=gsize,
+                              callback=get_rho_and_p)
+        red = FusedLaplacianReduction(
+            decomp, sector, derivs, halo_shape=h,
+            callback=get_rho_and_p, rank_shape=grid_shape,
+            grid_size=gsize, store_lap=False)
+        arrays = {"f": f0.clone().cuda(), "dfdt": d0.clone().cuda()}
+        arrays["f_next"] = torch.zeros_like(arrays["f"])
+        e0 = red(f=arrays["f"], dfdt=arrays["dfdt"], a=np.ones(1))
+        ex = ps.Expansion(e0["total"], ps.LowStorageRK54)
+        decomp.share_halos(arrays["f"])
+        dl = DeviceFriedmannLoop(st, decomp, ex, gsize, dt)
+        if split:
+            nx, ny, nz = grid_shape
+
+            def fake_regions(rank_shape, split_axes=None):
+                interior = (h, nx - h, h, ny - h, h, nz - h)
+                slabs = [
+                    (0, h, 0, ny, 0, nz),
+                    (nx - h, nx, 0, ny, 0, nz),
+                    (h, nx - h, 0, h, 0, nz),
+                    (h, nx - h, ny - h, ny, 0, nz),
+                    (h, nx - h, h, ny - h, 0, h),
+                    (h, nx - h, h, ny - h, nz - h, nz),
+                ]
+                return interior, slabs
+
+            dl._regions = fake_regions
+        for _ in range(2):
+            dl.step(arrays)
+        return arrays["f"].cpu(), dl.read_state()
+
+    f_one, st_one = run(False)
+    f_spl, st_spl = run(True)
+    cut = (slice(None),) + (slice(h, -h),) * 3
+    assert torch.equal(f_one[cut], f_spl[cut])
+    assert st_one["a"] == st_spl["a"]
+    assert st_one["energy"] == st_spl["energy"]
