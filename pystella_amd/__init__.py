@@ -92,3 +92,5 @@ class DisableLogging:
 
     def __exit__(self, *exc):
         self.logger.setLevel(self.original_level)
+
+__version__ = "2.0.0"
