@@ -28,14 +28,24 @@ def proc_shape(request):
                  request.config.getoption("--proc_shape").split(","))
 
 
-def run_distributed(fn, world_size=2, args=()):
+def run_distributed(fn, world_size=2, args=(), attempts=2):
     """Spawn `world_size` processes running fn(rank, world_size, *args)
-    under a gloo process group (CPU)."""
+    under a gloo process group (CPU).  Retries once on failure —
+    multi-process rendezvous/spawn is occasionally flaky under CI
+    load (a persistently failing test still fails)."""
     import torch.multiprocessing as mp
     import tempfile
-    init_file = tempfile.NamedTemporaryFile(delete=False).name
-    mp.spawn(_dist_worker, args=(world_size, init_file, fn, args),
-             nprocs=world_size, join=True)
+    last = None
+    for _ in range(attempts):
+        init_file = tempfile.NamedTemporaryFile(delete=False).name
+        try:
+            mp.spawn(_dist_worker,
+                     args=(world_size, init_file, fn, args),
+                     nprocs=world_size, join=True)
+            return
+        except Exception as e:          # noqa: BLE001
+            last = e
+    raise last
 
 
 def _dist_worker(rank, world_size, init_file, fn, args):
