@@ -6,7 +6,8 @@ pencil FFT with RCCL all-to-all transposes (reference used
 mpi4py-fft/FFTW on the host: pystella/fourier/dft.py:352-427).
 """
 
-from pystella_amd.fourier.dft import DFT, BaseDFT, fftfreq  # noqa: F401
+from pystella_amd.fourier.dft import (  # noqa: F401
+    DFT, BaseDFT, fftfreq, get_sliced_momenta, pDFT, pyclDFT)
 from pystella_amd.fourier.derivs import SpectralCollocator  # noqa: F401
 from pystella_amd.fourier.poisson import SpectralPoissonSolver  # noqa: F401
 from pystella_amd.fourier.projectors import Projector  # noqa: F401
@@ -28,3 +29,22 @@ def get_complex_dtype_with_matching_prec(dtype):
     if dtype in (np.dtype("float32"), np.dtype("complex64")):
         return np.dtype("complex64")
     return np.dtype("complex128")
+
+
+def gDFT(*args, **kwargs):
+    """Deprecated name kept for API parity (reference
+    fourier/dft.py:509 aliases its clFFT backend); here every
+    single-rank transform already runs the GPU path."""
+    from warnings import warn
+    warn("gDFT is deprecated; use DFT (or pyclDFT).",
+         DeprecationWarning, stacklevel=2)
+    return pyclDFT(*args, **kwargs)
+
+
+__all__ = [
+    "DFT", "BaseDFT", "fftfreq", "get_sliced_momenta", "pDFT",
+    "pyclDFT", "gDFT", "RayleighGenerator", "Projector",
+    "PowerSpectra", "SpectralCollocator", "SpectralPoissonSolver",
+    "get_real_dtype_with_matching_prec",
+    "get_complex_dtype_with_matching_prec",
+]
