@@ -80,6 +80,23 @@ class Field(Expr):
     def is_padded(self):
         return any(o == "h" for o in self.offset)
 
+    @property
+    def index_tuple(self):
+        """The fully-expanded subscript: each spatial index name offset
+        by the field's halo offset and static shift (reference
+        pystella/field/__init__.py:178-185)."""
+        out = []
+        for idx, off, sh in zip(self.indices, self.offset, self.shift):
+            term = Variable(idx) if isinstance(idx, str) else idx
+            if off == "h":
+                term = term + Variable("h")
+            elif off:
+                term = term + off
+            if sh:
+                term = term + sh
+            out.append(term)
+        return tuple(out)
+
     def copy(self, **kwargs):
         init = {a: getattr(self, a) for a in Field.init_args}
         init.update(kwargs)

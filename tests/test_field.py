@@ -139,3 +139,18 @@ def test_profiler_cpu():
         sum(range(1000))
     rep = prof.report()
     assert "work" in rep
+
+
+def test_index_tuple():
+    """Field.index_tuple: fully-expanded subscripts (reference
+    field/__init__.py:178-185)."""
+    import pystella_amd as ps
+    from pystella_amd.field import shift_fields
+    f = ps.Field("f", offset="h")
+    assert [str(t) for t in f.index_tuple] == \
+        ["i + h", "j + h", "k + h"]
+    g = ps.Field("g", offset=0)
+    assert [str(t) for t in g.index_tuple] == ["i", "j", "k"]
+    s = shift_fields(f, (1, 0, -2))
+    strs = [str(t) for t in s.index_tuple]
+    assert "1" in strs[0] and "-2" in strs[2].replace("(-2)", "-2")
