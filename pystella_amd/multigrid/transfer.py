@@ -13,7 +13,6 @@ variants apply the FAS corrections ``f2 -= R(f1)`` / ``f1 += I(f2)``.
 
 from __future__ import annotations
 
-from itertools import product
 
 __all__ = ["RestrictionBase", "FullWeighting", "Injection",
            "InterpolationBase", "LinearInterpolation",
@@ -32,22 +31,30 @@ class RestrictionBase:
     def __call__(self, queue=None, f1=None, f2=None):
         h = self.h
         n2 = tuple(s - 2 * h for s in f2.shape[-3:])
-        acc = None
-        for (a, ca), (b, cb), (c, cc) in product(
-                self.coefs.items(), self.coefs.items(),
-                self.coefs.items()):
-            w = ca * cb * cc
-            sl = (Ellipsis,
-                  slice(h + a, h + a + 2 * n2[0], 2),
-                  slice(h + b, h + b + 2 * n2[1], 2),
-                  slice(h + c, h + c + 2 * n2[2], 2))
-            term = w * f1[sl]
-            acc = term if acc is None else acc + term
+        # Separable tensor product: one strided pass per axis instead
+        # of len(coefs)^3 full-grid passes (the reference's loopy
+        # kernel reads each fine point once; the eager equivalent of
+        # that is per-axis factorization).  Axis order x, y, z: the
+        # stride-2 read only wastes HBM 32B-granule bandwidth on the
+        # contiguous (z) axis, so that pass runs last when the
+        # intermediate is already 4x smaller.
+        x = f1
+        items = sorted(self.coefs.items())
+        for k in range(3):
+            ax = x.ndim - 3 + k
+            sl = [slice(None)] * x.ndim
+            (a0, c0) = items[0]
+            sl[ax] = slice(h + a0, h + a0 + 2 * n2[k], 2)
+            acc = x[tuple(sl)].mul(c0)
+            for a, c in items[1:]:
+                sl[ax] = slice(h + a, h + a + 2 * n2[k], 2)
+                acc.add_(x[tuple(sl)], alpha=c)
+            x = acc
         interior = (Ellipsis,) + tuple(slice(h, h + n) for n in n2)
         if self.correct:
-            f2[interior] -= acc
+            f2[interior] -= x
         else:
-            f2[interior] = acc
+            f2[interior] = x
         return f2
 
 
@@ -85,31 +92,51 @@ class InterpolationBase:
         coefs = self.odd if parity else self.even
         return {(parity + a) // 2: c for a, c in coefs.items()}
 
+    def _axis_pass(self, x, k, h, n2k):
+        """Double axis k (coarse interior+halo extent -> fine interior
+        2*n2k) by the parity-dependent tensor-product rule."""
+        import torch
+        ax = x.ndim - 3 + k
+        halves = []
+        for p in (0, 1):
+            items = sorted(self._axis_coefs(p).items())
+            sl = [slice(None)] * x.ndim
+            (a0, c0) = items[0]
+            sl[ax] = slice(h + a0, h + a0 + n2k)
+            acc = x[tuple(sl)].mul(c0)
+            for a, c in items[1:]:
+                sl[ax] = slice(h + a, h + a + n2k)
+                acc.add_(x[tuple(sl)], alpha=c)
+            halves.append(acc)
+        if ax == x.ndim - 1:
+            # contiguous axis: interleave via stack+reshape so the
+            # write is a contiguous stream (a stride-2 scatter on the
+            # fastest axis would double the 32B-granule write traffic)
+            out = torch.stack(halves, dim=x.ndim)
+            return out.reshape(*x.shape[:-1], 2 * n2k)
+        shape = list(halves[0].shape)
+        shape[ax] = 2 * n2k
+        out = torch.empty(shape, dtype=x.dtype, device=x.device)
+        for p in (0, 1):
+            sl = [slice(None)] * x.ndim
+            sl[ax] = slice(p, 2 * n2k, 2)
+            out[tuple(sl)] = halves[p]
+        return out
+
     def __call__(self, queue=None, f1=None, f2=None):
         h = self.h
         n2 = tuple(s - 2 * h for s in f2.shape[-3:])
-        for px, py, pz in product((0, 1), repeat=3):
-            cx = self._axis_coefs(px)
-            cy = self._axis_coefs(py)
-            cz = self._axis_coefs(pz)
-            acc = None
-            for (a, ca), (b, cb), (c, cc) in product(
-                    cx.items(), cy.items(), cz.items()):
-                w = ca * cb * cc
-                sl = (Ellipsis,
-                      slice(h + a, h + a + n2[0]),
-                      slice(h + b, h + b + n2[1]),
-                      slice(h + c, h + c + n2[2]))
-                term = w * f2[sl]
-                acc = term if acc is None else acc + term
-            out = (Ellipsis,
-                   slice(h + px, h + 2 * n2[0], 2),
-                   slice(h + py, h + 2 * n2[1], 2),
-                   slice(h + pz, h + 2 * n2[2], 2))
-            if self.correct:
-                f1[out] += acc
-            else:
-                f1[out] = acc
+        # Separable per-axis doubling (3 passes) instead of 8 parity
+        # cases x len(coefs)^3 full-grid passes.  The contiguous axis
+        # is doubled FIRST, while the intermediate is smallest.
+        x = f2
+        for k in (2, 1, 0):
+            x = self._axis_pass(x, k, h, n2[k])
+        out = (Ellipsis,) + tuple(slice(h, h + 2 * n) for n in n2)
+        if self.correct:
+            f1[out] += x
+        else:
+            f1[out] = x
         return f1
 
 
