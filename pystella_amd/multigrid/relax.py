@@ -97,18 +97,52 @@ class RelaxationBase:
     def step(self, queue=None, **kwargs):
         self.stepper(**kwargs)
 
+    @property
+    def _star_operator(self):
+        """True when every stencil read in the operators is star-shaped
+        (at most one nonzero shift component) — then the smoothing
+        loop's intermediate halo shares only need faces, not corners."""
+        star = getattr(self, "_star_cached", None)
+        if star is None:
+            from pystella_amd.field import collect_fields
+            exprs = [lhs for (lhs, _rho) in self.lhs_dict.values()]
+            star = all(
+                sum(1 for s in fld.shift if s) <= 1
+                for fld in collect_fields(exprs))
+            self._star_cached = star
+        return star
+
+    def _smoother_share(self, decomp, arr, final=False):
+        """Halo share inside the smoothing loop.  Intermediate shares
+        of a star operator on a fully-local decomposition go through
+        the single-launch fused face wrap (backend/hip.py wrap_star, as
+        in the fused RK loop) — on the coarse levels the sequential
+        12-copy wrap is pure launch latency.  The FINAL share is always
+        a full (corner-propagating) share_halos: downstream restriction
+        reads 27-point corner halos."""
+        import torch
+        if (not final and self._star_operator
+                and all(p == 1 for p in decomp.proc_shape)
+                and len(set(self._h3)) == 1 and self._h3[0] > 0
+                and isinstance(arr, torch.Tensor) and arr.is_cuda):
+            from pystella_amd.backend.hip import wrap_star
+            wrap_star(arr, self._h3[0], (0, 1, 2))
+        else:
+            decomp.share_halos(arr)
+
     def __call__(self, decomp, queue=None, iterations=100, **kwargs):
         """Run ``iterations`` (rounded up to even) relaxation sweeps,
         ping-ponging each unknown with its ``tmp_`` array and sharing
         halos after every sweep (reference relax.py:164-200)."""
         kwargs.pop("solve_constraint", None)
         even_iterations = iterations + (iterations % 2)
-        for _ in range(even_iterations):
+        for it in range(even_iterations):
+            final = it == even_iterations - 1
             self.stepper(**kwargs)
             for name in self.unknown_names:
                 kwargs[name], kwargs["tmp_" + name] = \
                     kwargs["tmp_" + name], kwargs[name]
-                decomp.share_halos(kwargs[name])
+                self._smoother_share(decomp, kwargs[name], final=final)
 
     def get_error(self, queue=None, **kwargs):
         """L∞ and L2 norms of the residual per unknown
@@ -275,9 +309,12 @@ class RedBlackIterator(RelaxationBase):
                 _, start = decomp.get_rank_shape_start(decomp.grid_shape)
                 rb_off = float(sum(start) % 2)
             kwargs["rb_off"] = rb_off
-        for _ in range(iterations):
-            for stepper in self.color_steppers:
+        for it in range(iterations):
+            for ci, stepper in enumerate(self.color_steppers):
+                final = (it == iterations - 1
+                         and ci == len(self.color_steppers) - 1)
                 stepper(**{k: v for k, v in kwargs.items()
                            if not k.startswith("tmp_")})
                 for name in self.unknown_names:
-                    decomp.share_halos(kwargs[name])
+                    self._smoother_share(decomp, kwargs[name],
+                                         final=final)
