@@ -309,6 +309,8 @@ class RedBlackIterator(RelaxationBase):
                 _, start = decomp.get_rank_shape_start(decomp.grid_shape)
                 rb_off = float(sum(start) % 2)
             kwargs["rb_off"] = rb_off
+        if self._graph_call(decomp, iterations, kwargs):
+            return
         for it in range(iterations):
             for ci, stepper in enumerate(self.color_steppers):
                 final = (it == iterations - 1
@@ -318,3 +320,79 @@ class RedBlackIterator(RelaxationBase):
                 for name in self.unknown_names:
                     self._smoother_share(decomp, kwargs[name],
                                          final=final)
+
+    # -- hipGraph fast path --------------------------------------------
+    # The in-place two-color loop is launch-latency bound on coarse
+    # levels (each sweep = 2 stepper kernels + 2 fused wraps, all
+    # in-place with by-value scalars — no host work between launches).
+    # Capture the whole nu-iteration loop once per (level arrays,
+    # iterations) into a hipGraph and replay it as ONE launch.  The
+    # final corner-propagating share runs outside the graph.  Disable
+    # with PYSTELLA_MG_GRAPH=0.
+    def _graph_call(self, decomp, iterations, kwargs):
+        import os
+        import torch
+        if (os.environ.get("PYSTELLA_MG_GRAPH", "1") != "1"
+                or not self._star_operator
+                or not all(p == 1 for p in decomp.proc_shape)
+                or len(set(self._h3)) != 1 or self._h3[0] <= 0):
+            return False
+        tensors = {k: v for k, v in kwargs.items()
+                   if isinstance(v, torch.Tensor)}
+        if not tensors or not all(v.is_cuda for v in tensors.values()):
+            return False
+        scalars = tuple(sorted(
+            (k, tuple(np.asarray(v, dtype=float).reshape(-1).tolist()))
+            for k, v in kwargs.items()
+            if isinstance(v, (int, float, np.floating, np.ndarray))
+            and np.asarray(v).size >= 1))
+        key = (iterations, scalars, tuple(sorted(
+            (k, v.data_ptr(), tuple(v.shape))
+            for k, v in tensors.items())))
+        cache = getattr(self, "_graphs", None)
+        if cache is None:
+            cache = self._graphs = {}
+        graph = cache.get(key)
+        if graph is None:
+            from pystella_amd.backend.hip import wrap_star
+            step_kwargs = {k: v for k, v in kwargs.items()
+                           if not k.startswith("tmp_")}
+
+            def body():
+                for _ in range(iterations):
+                    for stepper in self.color_steppers:
+                        stepper(**step_kwargs)
+                        for name in self.unknown_names:
+                            wrap_star(kwargs[name], self._h3[0],
+                                      (0, 1, 2))
+            try:
+                # warm up on a side stream with CLONED state (compiles
+                # and caches every kernel so capture sees only
+                # launches, without mutating the real arrays)
+                warm_kwargs = {
+                    k: (v.clone() if isinstance(v, torch.Tensor) else v)
+                    for k, v in step_kwargs.items()}
+                s = torch.cuda.Stream()
+                s.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(s):
+                    for stepper in self.color_steppers:
+                        stepper(**warm_kwargs)
+                    for name in self.unknown_names:
+                        wrap_star(warm_kwargs[name], self._h3[0],
+                                  (0, 1, 2))
+                torch.cuda.current_stream().wait_stream(s)
+                torch.cuda.synchronize()
+                del warm_kwargs
+                graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(graph):
+                    body()
+            except Exception:
+                cache[key] = False
+                return False
+            cache[key] = graph
+        elif graph is False:
+            return False
+        graph.replay()
+        for name in self.unknown_names:
+            decomp.share_halos(kwargs[name])
+        return True

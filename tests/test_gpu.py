@@ -1353,3 +1353,58 @@ def test_smoother_star_wrap_bit_equality(n=32, h=1):
     # interiors + faces identical bit for bit; corners of the halo may
     # legitimately differ mid-loop but the FINAL share refreshes all
     assert torch.equal(fast, full)
+
+
+@requires_gpu
+def test_rbgs_graph_bit_equality(n=24, h=1):
+    """The hipGraph-captured RBGS smoothing loop replays bit-identically
+    to the eager loop (PYSTELLA_MG_GRAPH=0), across repeated calls
+    (graph cache reuse) and after state changes between calls."""
+    import os
+    from pystella_amd.field import Field, shift_fields
+    from pystella_amd.multigrid import RedBlackIterator
+
+    decomp = ps.DomainDecomposition((1, 1, 1), h, rank_shape=(n, n, n))
+    dx = (2 * np.pi / n,) * 3
+    f = Field("f", offset="h")
+    rho = Field("rho", offset="h")
+    lap = sum(
+        (shift_fields(f, tuple(s * int(m == d) for m in range(3)))
+         - 2 * f
+         + shift_fields(f, tuple(-s * int(m == d) for m in range(3))))
+        for d in range(3) for s in [1]) / var("dx")[0]**2
+    torch.manual_seed(29)
+    pad = (n + 2 * h,) * 3
+    rho_t = (torch.rand(pad, dtype=torch.float64, device="cuda") - 0.5)
+    decomp.share_halos(rho_t)
+
+    def run(graph):
+        old = os.environ.get("PYSTELLA_MG_GRAPH")
+        os.environ["PYSTELLA_MG_GRAPH"] = "1" if graph else "0"
+        try:
+            solver = RedBlackIterator(
+                decomp, {f: (lap, rho)}, halo_shape=h,
+                fixed_parameters=dict(omega=1.0))
+            ff = torch.zeros(pad, dtype=torch.float64, device="cuda")
+            out = []
+            for rep in range(3):   # repeated calls: capture then replays
+                solver(decomp, iterations=5, f=ff, rho=rho_t,
+                       dx=np.array(dx))
+                torch.cuda.synchronize()
+                out.append(ff.clone())
+            if graph:
+                assert any(g is not False
+                           for g in getattr(solver, "_graphs",
+                                            {}).values()), \
+                    "graph path did not engage"
+            return out
+        finally:
+            if old is None:
+                os.environ.pop("PYSTELLA_MG_GRAPH", None)
+            else:
+                os.environ["PYSTELLA_MG_GRAPH"] = old
+
+    eager = run(graph=False)
+    graphed = run(graph=True)
+    for a, b in zip(eager, graphed):
+        assert torch.equal(a, b)
