@@ -327,12 +327,15 @@ class RedBlackIterator(RelaxationBase):
     # in-place with by-value scalars — no host work between launches).
     # Capture the whole nu-iteration loop once per (level arrays,
     # iterations) into a hipGraph and replay it as ONE launch.  The
-    # final corner-propagating share runs outside the graph.  Disable
-    # with PYSTELLA_MG_GRAPH=0.
+    # final corner-propagating share runs outside the graph.
+    # Measured NEUTRAL at 1024^3 fp32 once the fused wrap landed
+    # (0.337 vs 0.335 s/V-cycle, profiles/r02_mg_graph_ab.txt) - the
+    # wrap fusion already removed the launch-latency tail - so this
+    # stays opt-in: PYSTELLA_MG_GRAPH=1.
     def _graph_call(self, decomp, iterations, kwargs):
         import os
         import torch
-        if (os.environ.get("PYSTELLA_MG_GRAPH", "1") != "1"
+        if (os.environ.get("PYSTELLA_MG_GRAPH", "0") != "1"
                 or not self._star_operator
                 or not all(p == 1 for p in decomp.proc_shape)
                 or len(set(self._h3)) != 1 or self._h3[0] <= 0):
