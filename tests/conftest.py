@@ -59,22 +59,47 @@ def _dist_worker(rank, world_size, init_file, fn, args):
         dist.destroy_process_group()
 
 
-def run_torchrun(cmd, cwd, log_dir, attempts=3, timeout=600):
-    """Run a torchrun command, retrying once on failure (multi-process
+def run_torchrun(cmd, cwd, log_dir, attempts=3, timeout=240):
+    """Run a torchrun command, retrying on failure (multi-process
     rendezvous is occasionally flaky under CI load); returns
-    (CompletedProcess, combined log text)."""
+    (CompletedProcess, combined log text).
+
+    Each RETRY rebinds --master-port to a freshly-probed free port:
+    the caller picks its port by bind-and-close, so the original can
+    be stolen in the close->rendezvous window (or still be held by an
+    orphan of a failed attempt) — retrying on the same port would then
+    fail all attempts identically."""
     import glob
+    import socket
     import subprocess
+    import time
     env = {k: v for k, v in __import__("os").environ.items()
            if k not in ("RANK", "LOCAL_RANK", "WORLD_SIZE",
                         "MASTER_ADDR", "MASTER_PORT", "LOCAL_WORLD_SIZE",
                         "GROUP_RANK", "TORCHELASTIC_RUN_ID")}
     out = None
-    for _ in range(attempts):
-        out = subprocess.run(cmd, capture_output=True, text=True,
-                             timeout=timeout, cwd=cwd, env=env)
+    logs = ""
+    for attempt in range(attempts):
+        c = list(cmd)
+        if attempt and "--master-port" in c:
+            with socket.socket() as s:
+                s.bind(("127.0.0.1", 0))
+                c[c.index("--master-port") + 1] = \
+                    str(s.getsockname()[1])
+        try:
+            out = subprocess.run(c, capture_output=True, text=True,
+                                 timeout=timeout, cwd=cwd, env=env)
+        except subprocess.TimeoutExpired as e:
+            out = subprocess.CompletedProcess(
+                c, returncode=-1,
+                stdout=(e.stdout or b"").decode(errors="replace")
+                if isinstance(e.stdout, bytes) else (e.stdout or ""),
+                stderr="torchrun attempt timed out")
+            time.sleep(2)
+            continue
         logs = "\n".join(open(f).read() for f in glob.glob(
             str(log_dir) + "/**/*.log", recursive=True))
         if out.returncode == 0:
             return out, logs
+        time.sleep(2)
     return out, logs
